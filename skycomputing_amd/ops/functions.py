@@ -1,0 +1,259 @@
+"""torch.autograd.Function wrappers over the HIP kernel library.
+
+Each Function's forward/backward launches hand-written gfx950 kernels on the
+current HIP stream via the C ABI in hiplib. Numerics contract: bf16 (or fp32)
+I/O, fp32 accumulation inside the kernels; tested against ops/eager.py in
+fp32 (tests/test_ops_gpu.py).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import hiplib
+from .hiplib import check, ptr
+
+_DT = {torch.float32: 0, torch.bfloat16: 1}
+
+_seed_gen = torch.Generator()
+_seed_gen.manual_seed(0x5EED)
+
+
+def set_dropout_seed(seed: int):
+    _seed_gen.manual_seed(seed)
+
+
+def _next_seed() -> int:
+    return int(torch.randint(0, 2**62, (1,), generator=_seed_gen).item())
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _dt(t: torch.Tensor) -> int:
+    try:
+        return _DT[t.dtype]
+    except KeyError:
+        raise TypeError(f"unsupported dtype {t.dtype} (fp32/bf16 only)") from None
+
+
+class LayerNormFn(torch.autograd.Function):
+    """Fused (residual-add +) LayerNorm, fwd+bwd in single-pass HIP kernels
+    (replaces the reference's optional apex FusedLayerNorm,
+    reference: scaelum/model/bert_layers.py:128-168)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps, residual):
+        lib = hiplib.require()
+        x = x.contiguous()
+        residual = residual.contiguous() if residual is not None else None
+        cols = x.shape[-1]
+        rows = x.numel() // cols
+        y = torch.empty_like(x)
+        mean = torch.empty(rows, dtype=torch.float32, device=x.device)
+        rstd = torch.empty(rows, dtype=torch.float32, device=x.device)
+        check(
+            lib.sky_layernorm_fwd(
+                _stream(), ptr(x), ptr(residual), ptr(weight), ptr(bias),
+                ptr(y), ptr(mean), ptr(rstd), rows, cols, eps, _dt(x),
+            ),
+            "sky_layernorm_fwd",
+        )
+        ctx.save_for_backward(x, weight, mean, rstd)
+        ctx.residual = residual
+        ctx.has_residual = residual is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        lib = hiplib.require()
+        x, weight, mean, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        cols = x.shape[-1]
+        rows = x.numel() // cols
+        dx = torch.empty_like(x)
+        dw32 = torch.zeros(cols, dtype=torch.float32, device=x.device)
+        db32 = torch.zeros(cols, dtype=torch.float32, device=x.device)
+        check(
+            lib.sky_layernorm_bwd(
+                _stream(), ptr(dy), ptr(x), ptr(ctx.residual), ptr(weight),
+                ptr(mean), ptr(rstd), ptr(dx), ptr(dw32), ptr(db32),
+                rows, cols, _dt(x),
+            ),
+            "sky_layernorm_bwd",
+        )
+        dw = dw32.to(weight.dtype)
+        db = db32.to(weight.dtype)
+        return dx, dw, db, None, (dx if ctx.has_residual else None)
+
+
+class BiasGeluFn(torch.autograd.Function):
+    """Fused bias + erf-GELU (reference: scaelum/model/bert_layers.py:21-44)."""
+
+    @staticmethod
+    def forward(ctx, x, bias):
+        lib = hiplib.require()
+        x = x.contiguous()
+        cols = x.shape[-1]
+        rows = x.numel() // cols
+        y = torch.empty_like(x)
+        check(
+            lib.sky_bias_gelu_fwd(_stream(), ptr(x), ptr(bias), ptr(y), rows, cols, _dt(x)),
+            "sky_bias_gelu_fwd",
+        )
+        ctx.save_for_backward(x, bias)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        lib = hiplib.require()
+        x, bias = ctx.saved_tensors
+        dy = dy.contiguous()
+        cols = x.shape[-1]
+        rows = x.numel() // cols
+        dx = torch.empty_like(x)
+        db32 = torch.zeros(cols, dtype=torch.float32, device=x.device)
+        check(
+            lib.sky_bias_gelu_bwd(
+                _stream(), ptr(dy), ptr(x), ptr(bias), ptr(dx), ptr(db32), rows, cols, _dt(x)
+            ),
+            "sky_bias_gelu_bwd",
+        )
+        return dx, db32.to(bias.dtype)
+
+
+class MaskedSoftmaxFn(torch.autograd.Function):
+    """Fused scale + additive-mask + row softmax over attention scores
+    [B, h, Sq, Sk]; mask [B, 1, 1, Sk] additive, broadcast over (h, Sq).
+    Replaces the reference's eager mask-add + softmax
+    (reference: scaelum/model/bert_layers.py:259-269)."""
+
+    @staticmethod
+    def forward(ctx, scores, mask, scale):
+        lib = hiplib.require()
+        scores = scores.contiguous()
+        B, h, Sq, Sk = scores.shape
+        probs = torch.empty_like(scores)
+        mask = mask.contiguous() if mask is not None else None
+        check(
+            lib.sky_masked_softmax_fwd(
+                _stream(), ptr(scores), ptr(mask), ptr(probs),
+                B, h, Sq, Sk, scale, 1.0, 0, _dt(scores),
+            ),
+            "sky_masked_softmax_fwd",
+        )
+        ctx.save_for_backward(probs)
+        ctx.scale = scale
+        return probs
+
+    @staticmethod
+    def backward(ctx, dp):
+        lib = hiplib.require()
+        (probs,) = ctx.saved_tensors
+        dp = dp.contiguous()
+        B, h, Sq, Sk = probs.shape
+        ds = torch.empty_like(probs)
+        check(
+            lib.sky_masked_softmax_bwd(
+                _stream(), ptr(dp), ptr(probs), ptr(ds),
+                B, h, Sq, Sk, ctx.scale, 1.0, 0, _dt(probs),
+            ),
+            "sky_masked_softmax_bwd",
+        )
+        return ds, None, None
+
+
+class DropoutFn(torch.autograd.Function):
+    """Dropout with counter-based RNG: the keep mask is regenerated from the
+    seed in backward, so no mask tensor is stored."""
+
+    @staticmethod
+    def forward(ctx, x, p):
+        lib = hiplib.require()
+        x = x.contiguous()
+        y = torch.empty_like(x)
+        seed = _next_seed()
+        keep = 1.0 - p
+        check(
+            lib.sky_dropout_fwd(_stream(), ptr(x), ptr(y), x.numel(), keep, seed, _dt(x)),
+            "sky_dropout_fwd",
+        )
+        ctx.seed = seed
+        ctx.keep = keep
+        ctx.meta = (x.numel(), x.dtype, x.device)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        lib = hiplib.require()
+        dy = dy.contiguous()
+        dx = torch.empty_like(dy)
+        check(
+            lib.sky_dropout_bwd(
+                _stream(), ptr(dy), ptr(dx), dy.numel(), ctx.keep, ctx.seed, _dt(dy)
+            ),
+            "sky_dropout_bwd",
+        )
+        return dx, None
+
+
+class EmbeddingFusedFn(torch.autograd.Function):
+    """Fused word+position+type gather, 3-way add and LayerNorm
+    (reference eager sequence: scaelum/model/bert_layers.py:191-212)."""
+
+    @staticmethod
+    def forward(ctx, input_ids, token_type_ids, position_ids,
+                word_emb, pos_emb, type_emb, ln_w, ln_b, eps):
+        lib = hiplib.require()
+        ids = input_ids.contiguous().view(-1).to(torch.int64)
+        tids = token_type_ids.contiguous().view(-1).to(torch.int64)
+        pids = position_ids.contiguous().view(-1).to(torch.int64)
+        rows = ids.numel()
+        cols = word_emb.shape[1]
+        y = torch.empty(
+            (*input_ids.shape, cols), dtype=word_emb.dtype, device=word_emb.device
+        )
+        mean = torch.empty(rows, dtype=torch.float32, device=y.device)
+        rstd = torch.empty(rows, dtype=torch.float32, device=y.device)
+        check(
+            lib.sky_embedding_fwd(
+                _stream(), ptr(ids), ptr(tids), ptr(pids),
+                ptr(word_emb), ptr(type_emb), ptr(pos_emb),
+                ptr(ln_w), ptr(ln_b), ptr(y), ptr(mean), ptr(rstd),
+                rows, cols, word_emb.shape[0], eps, _dt(y),
+            ),
+            "sky_embedding_fwd",
+        )
+        ctx.save_for_backward(ids, tids, pids, word_emb, pos_emb, type_emb, ln_w, mean, rstd)
+        ctx.eps = eps
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        lib = hiplib.require()
+        ids, tids, pids, word_emb, pos_emb, type_emb, ln_w, mean, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        cols = word_emb.shape[1]
+        rows = ids.numel()
+        dev = dy.device
+        dwe = torch.zeros(word_emb.shape, dtype=torch.float32, device=dev)
+        dte = torch.zeros(type_emb.shape, dtype=torch.float32, device=dev)
+        dpe = torch.zeros(pos_emb.shape, dtype=torch.float32, device=dev)
+        dlnw = torch.zeros(cols, dtype=torch.float32, device=dev)
+        dlnb = torch.zeros(cols, dtype=torch.float32, device=dev)
+        check(
+            lib.sky_embedding_bwd(
+                _stream(), ptr(dy), ptr(ids), ptr(tids), ptr(pids),
+                ptr(word_emb), ptr(type_emb), ptr(pos_emb), ptr(ln_w),
+                ptr(mean), ptr(rstd), ptr(dwe), ptr(dte), ptr(dpe),
+                ptr(dlnw), ptr(dlnb),
+                rows, cols, _dt(dy),
+            ),
+            "sky_embedding_bwd",
+        )
+        return (
+            None, None, None,
+            dwe.to(word_emb.dtype), dpe.to(pos_emb.dtype), dte.to(type_emb.dtype),
+            dlnw.to(ln_w.dtype), dlnb.to(ln_w.dtype), None,
+        )

@@ -1,0 +1,100 @@
+"""ctypes loader for the in-tree HIP kernel library (libskyhip.so).
+
+The kernels are hand-written CDNA4 (gfx950) HIP — see ops/hip/*.hip — built
+in-tree by ``skycomputing_amd.ops.build`` (driven from __graft_entry__.build())
+so the .so travels with the repo snapshot to GPU boxes.
+
+Design: a plain C ABI over raw device pointers + the current HIP stream,
+loaded with ctypes. This avoids libtorch C++ ABI coupling entirely; autograd
+integration lives in Python (ops/functions.py).
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+
+_LIB = None
+_LIB_PATH = os.path.join(os.path.dirname(os.path.abspath(__file__)), "hip", "libskyhip.so")
+_TRIED = False
+
+# every kernel entry point returns int (0 = ok, nonzero = hipError_t)
+_SIGNATURES: dict[str, list] = {}
+
+u64 = ctypes.c_uint64
+i64 = ctypes.c_int64
+i32 = ctypes.c_int32
+f32 = ctypes.c_float
+f64 = ctypes.c_double
+
+
+def _register_signatures(lib):
+    sigs = {
+        # layernorm: fwd saves mean/rstd (fp32 per row) for bwd
+        "sky_layernorm_fwd": [u64, u64, u64, u64, u64, u64, u64, u64, i64, i64, f32, i32],
+        #                     strm  x   res  w    b    y    mean rstd rows cols eps  dt
+        "sky_layernorm_bwd": [u64, u64, u64, u64, u64, u64, u64, u64, u64, u64, i64, i64, i32],
+        #                     strm  dy  x    res  w    mean rstd dx   dw   db   rows cols dt
+        "sky_bias_gelu_fwd": [u64, u64, u64, u64, i64, i64, i32],
+        #                     strm  x    b    y    rows cols dt
+        "sky_bias_gelu_bwd": [u64, u64, u64, u64, u64, u64, i64, i64, i32],
+        #                     strm  dy   x    b    dx   db   rows cols dt
+        "sky_masked_softmax_fwd": [u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, i32],
+        #                     strm scores mask probs B   h    Sq   Sk  scale keep seed dt
+        "sky_masked_softmax_bwd": [u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, i32],
+        #                     strm  dp  probs ds    B    h    Sq   Sk  scale keep seed dt
+        "sky_embedding_fwd": [u64] * 12 + [i64, i64, i64, f32, i32],
+        # strm ids tids pids wemb temb pemb lnw lnb y mean rstd | rows cols vocab eps dt
+        "sky_embedding_bwd": [u64] * 16 + [i64, i64, i32],
+        # strm dy ids tids pids wemb temb pemb lnw mean rstd dwe dte dpe dlnw dlnb | rows cols dt
+        "sky_sgd_step": [u64, u64, u64, u64, u64, i64, f32, f32, f32, i32, i32],
+        #                strm  p    g    m    mast n   lr  mom   wd   dt   has
+        "sky_detect_mem": [u64, u64],  # free_out, total_out (host ptrs)
+        "sky_gelu_dropout_fwd": [u64, u64, u64, i64, f32, u64, i32],
+        "sky_dropout_fwd": [u64, u64, u64, i64, f32, u64, i32],
+        #                   strm  x    y    n   keep seed dt
+        "sky_dropout_bwd": [u64, u64, u64, i64, f32, u64, i32],
+    }
+    for name, argtypes in sigs.items():
+        if hasattr(lib, name):
+            fn = getattr(lib, name)
+            fn.argtypes = argtypes
+            fn.restype = ctypes.c_int
+    return sigs
+
+
+def lib():
+    """Return the loaded CDLL or None if unavailable."""
+    global _LIB, _TRIED
+    if _LIB is None and not _TRIED:
+        _TRIED = True
+        if os.path.isfile(_LIB_PATH):
+            _LIB = ctypes.CDLL(_LIB_PATH)
+            _register_signatures(_LIB)
+    return _LIB
+
+
+def available() -> bool:
+    return lib() is not None
+
+
+def require():
+    l = lib()
+    if l is None:
+        raise RuntimeError(
+            f"skycomputing_amd HIP kernel library not found at {_LIB_PATH}. "
+            "Build it with `python -m skycomputing_amd.ops.build` (or __graft_entry__.build()). "
+            "Running CUDA tensors through the eager fallback is disabled so GPU runs "
+            "cannot silently lose the native path; set SKY_ALLOW_EAGER_GPU=1 to override."
+        )
+    return l
+
+
+def check(rc: int, name: str):
+    if rc != 0:
+        raise RuntimeError(f"{name} failed with hipError_t={rc}")
+
+
+def ptr(t) -> int:
+    """Device pointer of a tensor (0 for None)."""
+    return 0 if t is None else t.data_ptr()
