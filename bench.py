@@ -1,0 +1,173 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: 160-layer BERT (H=1024, A=16) training step,
+MNLI-shaped synthetic data, bsz=32 seq=128, bf16, pipeline-parallel over
+N MI355X GPUs (BASELINE.json metric: sec/iter).
+
+Run directly (N=1) or under torch.distributed.run with --nproc-per-node N
+(one rank per GPU over RCCL). Emits ONE JSON line from rank 0.
+
+Extra knobs beyond the driver contract:
+  --layers L            encoder layers (default 160)
+  --allocate MODE       even | dynamic | optimal (default even)
+  --microbatches M      pipeline microbatches (default: 1 if N==1 else 8)
+  --slowdowns CSV       per-rank injected compute slowdown factors
+  --stimulate           seeded synthetic heterogeneity for the allocator
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=int(os.environ.get("WORLD_SIZE", 1)))
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--layers", type=int, default=160)
+    p.add_argument("--batch", type=int, default=32)
+    p.add_argument("--seq", type=int, default=128)
+    p.add_argument("--allocate", default="even", choices=["even", "dynamic", "optimal"])
+    p.add_argument("--microbatches", type=int, default=0)
+    p.add_argument("--schedule", default="gpipe", choices=["gpipe", "sequential"])
+    p.add_argument("--slowdowns", default="")
+    p.add_argument("--stimulate", action="store_true")
+    p.add_argument("--dropout", type=float, default=0.1)
+    p.add_argument("--json-out", default="")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    from skycomputing_amd.dataset import SyntheticGlueDataset
+    from skycomputing_amd.dynamics import (
+        Allocator, DeviceBenchmarker, ModelBenchmarker, WorkerManager,
+    )
+    from skycomputing_amd.models import bert_pipeline_config
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import PartitionPlan, PipelineEngine, init_distributed
+    from skycomputing_amd.stimulator import Stimulator
+
+    use_cuda = torch.cuda.is_available()
+    comm = init_distributed()
+    world = comm.world_size
+    rank = comm.rank
+    dtype = torch.bfloat16 if use_cuda else torch.float32
+    torch.manual_seed(1234 + rank)
+
+    M = args.microbatches or (1 if world == 1 else 8)
+    assert args.batch % M == 0
+
+    bert_cfg = dict(
+        hidden_size=1024, num_attention_heads=16, intermediate_size=4096,
+        vocab_size=30522, max_position_embeddings=512,
+        hidden_dropout_prob=args.dropout, attention_probs_dropout_prob=args.dropout,
+    )
+    layer_cfgs = bert_pipeline_config(args.layers, bert_cfg)
+    L = len(layer_cfgs)
+
+    slowdowns = [float(s) for s in args.slowdowns.split(",") if s] or [0.0] * world
+    wm = WorkerManager.from_world(world, [dict(slowdown=s) for s in slowdowns])
+
+    # ---- allocation ----
+    if args.allocate == "even" and not args.stimulate:
+        base, rem = divmod(L, world)
+        bounds = [0]
+        for i in range(world):
+            bounds.append(bounds[-1] + base + (1 if i < rem else 0))
+        plan = PartitionPlan(
+            stage_ranks=list(range(world)),
+            ranges=[(bounds[i], bounds[i + 1]) for i in range(world)],
+        )
+    else:
+        stim = Stimulator(world) if args.stimulate else None
+        db = DeviceBenchmarker(comm, batch_size=args.batch, seq_len=args.seq,
+                               iterations=5, stimulator=stim)
+        dev_results = db.benchmark(wm)
+        mb = ModelBenchmarker(layer_cfgs, batch_size=args.batch // M, seq_len=args.seq)
+        model_results = mb.benchmark() if rank == 0 else None
+        if rank == 0:
+            workers = [dict(rank=r, **dev_results[r]) for r in range(world)]
+            alloc = Allocator(model_results["flops"], model_results["mem"], workers)
+            plan = alloc.allocate(args.allocate)
+            plan_d = plan.to_dict()
+        else:
+            plan_d = None
+        plan = PartitionPlan.from_dict(comm.broadcast_object(plan_d, src=0))
+
+    sd = slowdowns[rank] if rank < len(slowdowns) else 0.0
+    engine = PipelineEngine(
+        comm, layer_cfgs, plan,
+        loss_fn=lambda logits, labels: torch.nn.functional.cross_entropy(
+            logits.float(), labels
+        ),
+        dtype=dtype,
+        stage_kwargs=dict(record_forward_time=bool(sd > 0), slowdown=sd),
+    )
+    opt = FusedSGD(engine.parameters(), lr=1e-3)
+
+    ds = SyntheticGlueDataset(size=args.batch * 4, max_seq_length=args.seq, seed=7)
+    loader = torch.utils.data.DataLoader(ds, batch_size=args.batch, shuffle=False, drop_last=True)
+    batches = list(loader)
+
+    def step(i):
+        data, labels = batches[i % len(batches)]
+        opt.zero_grad(set_to_none=True)
+        engine.run_iteration(data, labels, num_microbatches=M, schedule=args.schedule)
+        opt.step()
+
+    for i in range(args.warmup):
+        step(i)
+    comm.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    comm.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    elapsed = max(comm.all_gather_object(elapsed))
+    sec_per_iter = elapsed / args.steps
+
+    if rank == 0:
+        result = {
+            "metric": "sec_per_iter_bert160_mnli_bsz32",
+            "value": sec_per_iter,
+            "unit": "sec/iter",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": sec_per_iter * 1e3,
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_cuda else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"bert-{args.layers}L-H1024-A16",
+                "global_batch": args.batch,
+                "seq_len": args.seq,
+                "parallelism": f"pp{world}",
+                "microbatches": M,
+                "schedule": args.schedule,
+                "allocate": args.allocate,
+                "dropout": args.dropout,
+                "num_pipeline_layers": L,
+            },
+        }
+        line = json.dumps(result)
+        print(line, flush=True)
+        if args.json_out:
+            with open(args.json_out, "w") as f:
+                f.write(line + "\n")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,75 @@
+// Multi-tensor fused SGD for gfx950.
+//
+// Replaces per-parameter eager SGD (reference: DistributedOptimizer's local
+// SGD steps, experiment/launch.py:152-156): ONE kernel launch updates every
+// parameter of the stage. The host packs per-slab descriptors
+// {param_ptr, grad_ptr, master_ptr, momentum_ptr, count} into a device
+// int64 array (built once; pointers are stable after the first backward);
+// each block processes one slab. With fp32 master weights the update runs
+// on the master and the bf16 param is refreshed in the same pass.
+//
+// memutil: sky_detect_mem wraps hipMemGetInfo (replacing the reference's
+// nvidia-smi subprocess probe, module_wrapper.py:210-219).
+
+#include "common.h"
+
+struct SgdDesc {
+  uint64_t p, g, m, mom;
+  int64_t n;
+  int64_t pad;
+};
+
+template <int DT, int BLOCK, bool MASTER, bool MOM>
+__global__ __launch_bounds__(BLOCK) void sgd_kernel(
+    const SgdDesc* __restrict__ descs, float lr, float momentum, float wd) {
+  const SgdDesc d = descs[blockIdx.x];
+  void* p = (void*)d.p;
+  const void* g = (const void*)d.g;
+  float* master = (float*)d.m;
+  float* mbuf = (float*)d.mom;
+  for (int64_t i = threadIdx.x; i < d.n; i += BLOCK) {
+    float gv = load_elem<DT>(g, i);
+    float pv = MASTER ? master[i] : load_elem<DT>(p, i);
+    if (wd != 0.f) gv += wd * pv;
+    if (MOM) {
+      float b = mbuf[i] * momentum + gv;
+      mbuf[i] = b;
+      gv = b;
+    }
+    pv -= lr * gv;
+    if (MASTER) master[i] = pv;
+    store_elem<DT>(p, i, pv);
+  }
+}
+
+SKY_EXPORT int sky_sgd_step(uint64_t stream, uint64_t descs, int64_t n_descs,
+                            float lr, float momentum, float wd, int dt,
+                            int flags) {
+  constexpr int BLOCK = 256;
+  const bool has_master = flags & 1;
+  const bool has_mom = flags & 2;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid((unsigned)n_descs);
+#define SGD(DT, MA, MO)                                                      \
+  hipLaunchKernelGGL((sgd_kernel<DT, BLOCK, MA, MO>), grid, dim3(BLOCK), 0,  \
+                     s, (const SgdDesc*)descs, lr, momentum, wd)
+  if (dt == DT_F32) {
+    if (has_master) { if (has_mom) SGD(DT_F32, true, true); else SGD(DT_F32, true, false); }
+    else            { if (has_mom) SGD(DT_F32, false, true); else SGD(DT_F32, false, false); }
+  } else {
+    if (has_master) { if (has_mom) SGD(DT_BF16, true, true); else SGD(DT_BF16, true, false); }
+    else            { if (has_mom) SGD(DT_BF16, false, true); else SGD(DT_BF16, false, false); }
+  }
+#undef SGD
+  LAUNCH_CHECK();
+  return 0;
+}
+
+SKY_EXPORT int sky_detect_mem(uint64_t free_out, uint64_t total_out) {
+  size_t f = 0, t = 0;
+  hipError_t e = hipMemGetInfo(&f, &t);
+  if (e != hipSuccess) return (int)e;
+  *(uint64_t*)free_out = (uint64_t)f;
+  *(uint64_t*)total_out = (uint64_t)t;
+  return 0;
+}

@@ -47,10 +47,9 @@ def _register_signatures(lib):
         # strm ids tids pids wemb temb pemb lnw lnb y mean rstd | rows cols vocab eps dt
         "sky_embedding_bwd": [u64] * 16 + [i64, i64, i32],
         # strm dy ids tids pids wemb temb pemb lnw mean rstd dwe dte dpe dlnw dlnb | rows cols dt
-        "sky_sgd_step": [u64, u64, u64, u64, u64, i64, f32, f32, f32, i32, i32],
-        #                strm  p    g    m    mast n   lr  mom   wd   dt   has
+        "sky_sgd_step": [u64, u64, i64, f32, f32, f32, i32, i32],
+        #                strm descs n   lr   mom  wd   dt   flags
         "sky_detect_mem": [u64, u64],  # free_out, total_out (host ptrs)
-        "sky_gelu_dropout_fwd": [u64, u64, u64, i64, f32, u64, i32],
         "sky_dropout_fwd": [u64, u64, u64, i64, f32, u64, i32],
         #                   strm  x    y    n   keep seed dt
         "sky_dropout_bwd": [u64, u64, u64, i64, f32, u64, i32],

@@ -1,0 +1,104 @@
+"""Single-GPU end-to-end tests on MI355X (bf16, HIP kernels, full model)."""
+
+from __future__ import annotations
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _require_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from skycomputing_amd.ops import hiplib
+
+    assert hiplib.available()
+
+
+def test_bert_stage_trains_bf16():
+    """Loss decreases over a few steps of the full fused-op bf16 path."""
+    torch.manual_seed(0)
+    from skycomputing_amd.builder import build_module_from_cfg
+    from skycomputing_amd.models import bert_pipeline_config
+    from skycomputing_amd.optim import FusedSGD
+
+    cfgs = bert_pipeline_config(
+        2, dict(hidden_size=512, num_attention_heads=8, intermediate_size=2048,
+                vocab_size=5000, hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0)
+    )
+    stage = build_module_from_cfg(cfgs, device="cuda:0", dtype=torch.bfloat16)
+    opt = FusedSGD(stage.parameters(), lr=1e-2)
+    ids = torch.randint(0, 5000, (16, 64), device="cuda")
+    tids = torch.zeros_like(ids)
+    mask = torch.ones_like(ids)
+    labels = torch.randint(0, 3, (16,), device="cuda")
+    losses = []
+    for _ in range(12):
+        opt.zero_grad()
+        logits = stage(ids, tids, mask)
+        loss = torch.nn.functional.cross_entropy(logits.float(), labels)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0] * 0.8, losses
+
+
+def test_bf16_step_matches_fp32_reference():
+    """One training step of a small BERT stage in bf16 (HIP kernels) tracks
+    the fp32 eager reference within bf16 tolerance."""
+    torch.manual_seed(1)
+    from skycomputing_amd.builder import build_module_from_cfg
+
+    cfgs_kwargs = dict(hidden_size=256, num_attention_heads=4, intermediate_size=1024,
+                       vocab_size=2000, hidden_dropout_prob=0.0,
+                       attention_probs_dropout_prob=0.0)
+    from skycomputing_amd.models import bert_pipeline_config
+
+    cfgs = bert_pipeline_config(1, cfgs_kwargs)
+    torch.manual_seed(42)
+    gpu = build_module_from_cfg(cfgs, device="cuda:0", dtype=torch.bfloat16)
+    torch.manual_seed(42)
+    cpu = build_module_from_cfg(cfgs, device="cpu")
+    ids = torch.randint(0, 2000, (8, 32))
+    tids = torch.zeros_like(ids)
+    mask = torch.ones_like(ids)
+    lg = gpu(ids.cuda(), tids.cuda(), mask.cuda())
+    lc = cpu(ids, tids, mask)
+    assert torch.allclose(lg.float().cpu(), lc, atol=0.1, rtol=0.1), (
+        (lg.float().cpu() - lc).abs().max()
+    )
+    labels = torch.randint(0, 3, (8,))
+    torch.nn.functional.cross_entropy(lg.float(), labels.cuda()).backward()
+    torch.nn.functional.cross_entropy(lc.float(), labels).backward()
+    # compare a representative param grad (embeddings LN weight)
+    g_gpu = gpu.module[0].layer_norm.weight.grad.float().cpu()
+    g_cpu = cpu.module[0].layer_norm.weight.grad
+    assert torch.allclose(g_gpu, g_cpu, atol=0.05, rtol=0.1), (g_gpu - g_cpu).abs().max()
+
+
+def test_device_benchmark_and_detect_mem():
+    from skycomputing_amd.builder import build_module_from_cfg
+    from skycomputing_amd.dynamics.benchmarker import default_bert_probe_cfg
+
+    stage = build_module_from_cfg(
+        default_bert_probe_cfg(512), device="cuda:0", dtype=torch.bfloat16
+    )
+    free = stage.detect_mem()
+    assert free > 1 << 30  # at least 1 GB free on a 288 GB part
+
+    from skycomputing_amd.ops import hiplib
+    import ctypes
+
+    f = ctypes.c_uint64()
+    t = ctypes.c_uint64()
+    rc = hiplib.lib().sky_detect_mem(ctypes.addressof(f), ctypes.addressof(t))
+    assert rc == 0 and t.value > 200 * (1 << 30)  # 288 GB part
+
+
+def test_graft_smoke():
+    import __graft_entry__
+
+    __graft_entry__.smoke()
