@@ -76,7 +76,11 @@ class LinearActivation(nn.Module):
 
 class BertSelfAttention(nn.Module):
     """Multi-head self-attention core
-    (reference: scaelum/model/bert_layers.py:215-275)."""
+    (reference: scaelum/model/bert_layers.py:215-275).
+
+    MI355X design: ONE fused QKV GEMM [B*S,H]x[H,3H] instead of three
+    H x H GEMMs (bigger tiles fill the 256 CUs; one launch instead of
+    three) — SURVEY.md §2c row 1's fused-QKV option."""
 
     def __init__(self, config: BertConfig):
         super().__init__()
@@ -85,21 +89,18 @@ class BertSelfAttention(nn.Module):
         self.num_heads = config.num_attention_heads
         self.head_dim = config.hidden_size // config.num_attention_heads
         H = config.hidden_size
-        self.query = nn.Linear(H, H)
-        self.key = nn.Linear(H, H)
-        self.value = nn.Linear(H, H)
+        self.qkv = nn.Linear(H, 3 * H)
         self.dropout_p = config.attention_probs_dropout_prob
 
-    def _split_heads(self, x):
-        B, S, H = x.shape
-        return x.view(B, S, self.num_heads, self.head_dim).permute(0, 2, 1, 3)
-
     def forward(self, hidden, ext_mask):
-        q = self._split_heads(self.query(hidden))
-        k = self._split_heads(self.key(hidden))
-        v = self._split_heads(self.value(hidden))
+        B, S, H = hidden.shape
+        h, d = self.num_heads, self.head_dim
+        qkv = self.qkv(hidden).view(B, S, 3, h, d)
+        # strided views; batched GEMMs consume them without materializing
+        q = qkv[:, :, 0].permute(0, 2, 1, 3)
+        k = qkv[:, :, 1].permute(0, 2, 1, 3)
+        v = qkv[:, :, 2].permute(0, 2, 1, 3)
         ctx = ops.attention_context(q, k, v, ext_mask, self.dropout_p, self.training)
-        B, h, S, d = ctx.shape
         return ctx.permute(0, 2, 1, 3).reshape(B, S, h * d)
 
 

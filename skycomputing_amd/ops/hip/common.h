@@ -39,6 +39,39 @@ DEV ushort_t f32_to_bf16(float f) {
 // dtype tags matching hiplib._DT
 enum { DT_F32 = 0, DT_BF16 = 1 };
 
+// 8-element vector I/O (16 B/lane for bf16 — the coalescing sweet spot,
+// guide G13; fp32 goes through two float4). Index i8 is in units of 8
+// elements; pointers must be 16-byte aligned at those offsets.
+template <int DT> struct Vec8;
+template <> struct Vec8<DT_BF16> {
+  static DEV void load(const void* p, int64_t i8, float f[8]) {
+    ushort8_t v = ((const ushort8_t*)p)[i8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) f[j] = bf16_to_f32(v[j]);
+  }
+  static DEV void store(void* p, int64_t i8, const float f[8]) {
+    ushort8_t v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = f32_to_bf16(f[j]);
+    ((ushort8_t*)p)[i8] = v;
+  }
+};
+template <> struct Vec8<DT_F32> {
+  static DEV void load(const void* p, int64_t i8, float f[8]) {
+    float4_t a = ((const float4_t*)p)[i8 * 2];
+    float4_t b = ((const float4_t*)p)[i8 * 2 + 1];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) { f[j] = a[j]; f[4 + j] = b[j]; }
+  }
+  static DEV void store(void* p, int64_t i8, const float f[8]) {
+    float4_t a, b;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) { a[j] = f[j]; b[j] = f[4 + j]; }
+    ((float4_t*)p)[i8 * 2] = a;
+    ((float4_t*)p)[i8 * 2 + 1] = b;
+  }
+};
+
 // generic element load/store through a dtype tag (scalar path)
 template <int DT> DEV float load_elem(const void* p, int64_t i);
 template <> DEV float load_elem<DT_F32>(const void* p, int64_t i) {

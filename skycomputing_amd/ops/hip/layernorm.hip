@@ -2,20 +2,73 @@
 //
 // Replaces the reference's optional apex FusedLayerNormAffineFunction
 // (reference: scaelum/model/bert_layers.py:128-168) and fuses the
-// preceding residual add + dropout-output add pattern
-// (bert_layers.py:283-288,321-326) into the same pass.
+// preceding residual add (bert_layers.py:283-288,321-326) into the pass.
 //
-// Forward: one 256-thread block per row-group; per-row mean/var accumulated
-// in fp32 via wave shuffles + LDS cross-wave reduce; saves mean/rstd (fp32)
-// for backward. Memory-bound: bf16 loads vectorized 8-wide.
-// Backward: one block per ROWS_PER_WG rows; dweight/dbias accumulated in
-// registers across the block's rows, then one fp32 atomicAdd per column per
-// block (guide Guideline 12: partial-reduce before atomics).
+// Fast path (cols % 8 == 0, cols <= 4096): WAVE-PER-ROW, 16 B/lane
+// vectorized loads, values register-cached across the stats/normalize
+// passes, reductions via 64-lane shuffles only — no LDS, no barriers.
+// Backward splits into a wave-per-row dx kernel plus a column-parallel
+// dweight/dbias reduction kernel (coalesced down-column traversal, few
+// atomics) instead of row-serial LDS accumulation.
+// Generic fallback (odd cols): block-per-row scalar kernels.
 
 #include "common.h"
 
-// ---------------- forward ----------------
+#define LN_MAXCH 8  // max chunks of 512 elements -> cols <= 4096 fast path
 
+// ---------------- forward (vectorized, wave per row) ----------------
+
+template <int DT, bool HAS_RES, int CH>
+__global__ __launch_bounds__(256) void ln_fwd_vec_kernel(
+    const void* __restrict__ x, const void* __restrict__ res,
+    const void* __restrict__ w, const void* __restrict__ b,
+    void* __restrict__ y, float* __restrict__ mean_out,
+    float* __restrict__ rstd_out, int64_t rows, int64_t cols, float eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int64_t cols8 = cols / 8;
+  float v[CH][8];
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wid; row < rows;
+       row += (int64_t)gridDim.x * 4) {
+    const int64_t base8 = row * cols8;
+    float s = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int c = 0; c < CH; ++c) {
+      const int64_t i8 = (int64_t)c * WAVE + lane;
+      if (i8 < cols8) {
+        Vec8<DT>::load(x, base8 + i8, v[c]);
+        if (HAS_RES) {
+          float r[8];
+          Vec8<DT>::load(res, base8 + i8, r);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) v[c][j] += r[j];
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) { s += v[c][j]; s2 += v[c][j] * v[c][j]; }
+      }
+    }
+    s = wave_sum(s);
+    s2 = wave_sum(s2);
+    const float mean = s / (float)cols;
+    const float rstd = rsqrtf(fmaxf(s2 / (float)cols - mean * mean, 0.f) + eps);
+    if (lane == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
+#pragma unroll
+    for (int c = 0; c < CH; ++c) {
+      const int64_t i8 = (int64_t)c * WAVE + lane;
+      if (i8 < cols8) {
+        float wv[8], bv[8], o[8];
+        Vec8<DT>::load(w, i8, wv);
+        Vec8<DT>::load(b, i8, bv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o[j] = (v[c][j] - mean) * rstd * wv[j] + bv[j];
+        Vec8<DT>::store(y, base8 + i8, o);
+      }
+    }
+  }
+}
+
+// generic scalar fallback (block per row)
 template <int DT, int BLOCK, bool HAS_RES>
 __global__ __launch_bounds__(BLOCK) void ln_fwd_kernel(
     const void* __restrict__ x, const void* __restrict__ res,
@@ -35,37 +88,57 @@ __global__ __launch_bounds__(BLOCK) void ln_fwd_kernel(
     s = block_sum<BLOCK>(s, lds);
     s2 = block_sum<BLOCK>(s2, lds);
     const float mean = s / (float)cols;
-    float var = s2 / (float)cols - mean * mean;
-    const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
-    if (threadIdx.x == 0) {
-      mean_out[row] = mean;
-      rstd_out[row] = rstd;
-    }
+    const float rstd = rsqrtf(fmaxf(s2 / (float)cols - mean * mean, 0.f) + eps);
+    if (threadIdx.x == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
     for (int64_t c = threadIdx.x; c < cols; c += BLOCK) {
       float v = load_elem<DT>(x, base + c);
       if (HAS_RES) v += load_elem<DT>(res, base + c);
-      float wv = load_elem<DT>(w, c), bv = load_elem<DT>(b, c);
-      store_elem<DT>(y, base + c, (v - mean) * rstd * wv + bv);
+      store_elem<DT>(y, base + c,
+                     (v - mean) * rstd * load_elem<DT>(w, c) + load_elem<DT>(b, c));
     }
   }
+}
+
+static inline bool ln_fast_ok(int64_t cols) {
+  return (cols % 8 == 0) && cols <= 64 * 8 * LN_MAXCH;
 }
 
 SKY_EXPORT int sky_layernorm_fwd(uint64_t stream, uint64_t x, uint64_t res,
                                  uint64_t w, uint64_t b, uint64_t y,
                                  uint64_t mean, uint64_t rstd, int64_t rows,
                                  int64_t cols, float eps, int dt) {
-  constexpr int BLOCK = 256;
-  dim3 grid((unsigned)(rows < 4096 ? rows : 4096));
   hipStream_t s = (hipStream_t)stream;
   bool has_res = res != 0;
-#define LNF(DT, HR)                                                         \
-  hipLaunchKernelGGL((ln_fwd_kernel<DT, BLOCK, HR>), grid, dim3(BLOCK), 0, s, \
-                     (const void*)x, (const void*)res, (const void*)w,      \
-                     (const void*)b, (void*)y, (float*)mean, (float*)rstd,  \
+  if (ln_fast_ok(cols)) {
+    unsigned grid = (unsigned)((rows + 3) / 4);
+    if (grid > 8192u) grid = 8192u;
+    const int ch = (int)((cols / 8 + WAVE - 1) / WAVE);
+#define LNFV(DT, HR, CH)                                                      \
+  hipLaunchKernelGGL((ln_fwd_vec_kernel<DT, HR, CH>), dim3(grid), dim3(256),  \
+                     0, s, (const void*)x, (const void*)res, (const void*)w,  \
+                     (const void*)b, (void*)y, (float*)mean, (float*)rstd,    \
                      rows, cols, eps)
-  if (dt == DT_F32) { if (has_res) LNF(DT_F32, true); else LNF(DT_F32, false); }
-  else              { if (has_res) LNF(DT_BF16, true); else LNF(DT_BF16, false); }
+#define LNFV_CH(DT, HR)                                                       \
+  do {                                                                        \
+    if (ch <= 2) LNFV(DT, HR, 2);                                             \
+    else if (ch <= 4) LNFV(DT, HR, 4);                                        \
+    else LNFV(DT, HR, 8);                                                     \
+  } while (0)
+    if (dt == DT_F32) { if (has_res) LNFV_CH(DT_F32, true); else LNFV_CH(DT_F32, false); }
+    else              { if (has_res) LNFV_CH(DT_BF16, true); else LNFV_CH(DT_BF16, false); }
+#undef LNFV_CH
+#undef LNFV
+  } else {
+    dim3 grid((unsigned)(rows < 4096 ? rows : 4096));
+#define LNF(DT, HR)                                                           \
+  hipLaunchKernelGGL((ln_fwd_kernel<DT, 256, HR>), grid, dim3(256), 0, s,     \
+                     (const void*)x, (const void*)res, (const void*)w,        \
+                     (const void*)b, (void*)y, (float*)mean, (float*)rstd,    \
+                     rows, cols, eps)
+    if (dt == DT_F32) { if (has_res) LNF(DT_F32, true); else LNF(DT_F32, false); }
+    else              { if (has_res) LNF(DT_BF16, true); else LNF(DT_BF16, false); }
 #undef LNF
+  }
   LAUNCH_CHECK();
   return 0;
 }
@@ -75,43 +148,103 @@ SKY_EXPORT int sky_layernorm_fwd(uint64_t stream, uint64_t x, uint64_t res,
 // dx = rstd * (dxhat - mean_c(dxhat) - xhat * mean_c(dxhat * xhat))
 // dw[c] += sum_r dy * xhat ; db[c] += sum_r dy
 
-template <int DT, int BLOCK, bool HAS_RES, int ROWS_PER_WG>
-__global__ __launch_bounds__(BLOCK) void ln_bwd_kernel(
+template <int DT, bool HAS_RES, int CH>
+__global__ __launch_bounds__(256) void ln_bwd_dx_vec_kernel(
     const void* __restrict__ dy, const void* __restrict__ x,
     const void* __restrict__ res, const void* __restrict__ w,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    void* __restrict__ dx, float* __restrict__ dw, float* __restrict__ db,
-    int64_t rows, int64_t cols) {
-  // single dynamic LDS region: [cols] dw partials, [cols] db partials,
-  // [BLOCK/WAVE] reduce scratch (one __shared__ object, 16B-aligned base —
-  // guide Guideline 17)
-  extern __shared__ __attribute__((aligned(16))) float dwdb[];
-  float* dw_part = dwdb;
-  float* db_part = dwdb + cols;
-  float* lds = dwdb + 2 * cols;
-  for (int64_t c = threadIdx.x; c < cols; c += BLOCK) {
-    dw_part[c] = 0.f;
-    db_part[c] = 0.f;
+    void* __restrict__ dx, int64_t rows, int64_t cols) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int64_t cols8 = cols / 8;
+  float xh[CH][8], dxh[CH][8];
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wid; row < rows;
+       row += (int64_t)gridDim.x * 4) {
+    const int64_t base8 = row * cols8;
+    const float mu = mean[row], rs = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int c = 0; c < CH; ++c) {
+      const int64_t i8 = (int64_t)c * WAVE + lane;
+      if (i8 < cols8) {
+        float xv[8], dyv[8], wv[8];
+        Vec8<DT>::load(x, base8 + i8, xv);
+        if (HAS_RES) {
+          float r[8];
+          Vec8<DT>::load(res, base8 + i8, r);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) xv[j] += r[j];
+        }
+        Vec8<DT>::load(dy, base8 + i8, dyv);
+        Vec8<DT>::load(w, i8, wv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          xh[c][j] = (xv[j] - mu) * rs;
+          dxh[c][j] = dyv[j] * wv[j];
+          s1 += dxh[c][j];
+          s2 += dxh[c][j] * xh[c][j];
+        }
+      }
+    }
+    s1 = wave_sum(s1) / (float)cols;
+    s2 = wave_sum(s2) / (float)cols;
+#pragma unroll
+    for (int c = 0; c < CH; ++c) {
+      const int64_t i8 = (int64_t)c * WAVE + lane;
+      if (i8 < cols8) {
+        float o[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o[j] = rs * (dxh[c][j] - s1 - xh[c][j] * s2);
+        Vec8<DT>::store(dx, base8 + i8, o);
+      }
+    }
   }
-  __syncthreads();
+}
 
-  const int64_t group = blockIdx.x;
-  const int64_t row0 = group * ROWS_PER_WG;
-  for (int64_t row = row0; row < rows && row < row0 + ROWS_PER_WG; ++row) {
+// column-parallel dw/db: thread t owns column col0+t; walks a row-slab down
+// (coalesced: consecutive threads -> consecutive columns), one atomicAdd
+// per column per slab.
+template <int DT, bool HAS_RES, int BLOCK>
+__global__ __launch_bounds__(BLOCK) void ln_bwd_wb_kernel(
+    const void* __restrict__ dy, const void* __restrict__ x,
+    const void* __restrict__ res, const float* __restrict__ mean,
+    const float* __restrict__ rstd, float* __restrict__ dw,
+    float* __restrict__ db, int64_t rows, int64_t cols, int64_t rows_per_slab) {
+  const int64_t col = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  if (col >= cols) return;
+  const int64_t r0 = (int64_t)blockIdx.y * rows_per_slab;
+  const int64_t r1 = min(rows, r0 + rows_per_slab);
+  float sw = 0.f, sb = 0.f;
+  for (int64_t r = r0; r < r1; ++r) {
+    float xv = load_elem<DT>(x, r * cols + col);
+    if (HAS_RES) xv += load_elem<DT>(res, r * cols + col);
+    float dyv = load_elem<DT>(dy, r * cols + col);
+    sw += dyv * (xv - mean[r]) * rstd[r];
+    sb += dyv;
+  }
+  atomicAdd(&dw[col], sw);
+  atomicAdd(&db[col], sb);
+}
+
+// generic scalar fallback dx (block per row)
+template <int DT, int BLOCK, bool HAS_RES>
+__global__ __launch_bounds__(BLOCK) void ln_bwd_dx_kernel(
+    const void* __restrict__ dy, const void* __restrict__ x,
+    const void* __restrict__ res, const void* __restrict__ w,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    void* __restrict__ dx, int64_t rows, int64_t cols) {
+  __shared__ float lds[BLOCK / WAVE > 2 ? BLOCK / WAVE : 2];
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const int64_t base = row * cols;
     const float mu = mean[row], rs = rstd[row];
     float s1 = 0.f, s2 = 0.f;
     for (int64_t c = threadIdx.x; c < cols; c += BLOCK) {
       float xv = load_elem<DT>(x, base + c);
       if (HAS_RES) xv += load_elem<DT>(res, base + c);
-      float xhat = (xv - mu) * rs;
-      float dyv = load_elem<DT>(dy, base + c);
-      float wv = load_elem<DT>(w, c);
-      float dxhat = dyv * wv;
+      float dxhat = load_elem<DT>(dy, base + c) * load_elem<DT>(w, c);
       s1 += dxhat;
-      s2 += dxhat * xhat;
-      dw_part[c] += dyv * xhat;
-      db_part[c] += dyv;
+      s2 += dxhat * (xv - mu) * rs;
     }
     s1 = block_sum<BLOCK>(s1, lds) / (float)cols;
     s2 = block_sum<BLOCK>(s2, lds) / (float)cols;
@@ -122,11 +255,6 @@ __global__ __launch_bounds__(BLOCK) void ln_bwd_kernel(
       float dxhat = load_elem<DT>(dy, base + c) * load_elem<DT>(w, c);
       store_elem<DT>(dx, base + c, rs * (dxhat - s1 - xhat * s2));
     }
-    __syncthreads();
-  }
-  for (int64_t c = threadIdx.x; c < cols; c += BLOCK) {
-    atomicAdd(&dw[c], dw_part[c]);
-    atomicAdd(&db[c], db_part[c]);
   }
 }
 
@@ -135,22 +263,55 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
                                  uint64_t rstd, uint64_t dx, uint64_t dw,
                                  uint64_t db, int64_t rows, int64_t cols,
                                  int dt) {
-  constexpr int BLOCK = 256;
-  constexpr int RPW = 8;
-  size_t lds_bytes = (2 * cols + BLOCK / WAVE) * sizeof(float);
-  if (lds_bytes > 64 * 1024) return (int)hipErrorInvalidValue;
-  dim3 grid((unsigned)((rows + RPW - 1) / RPW));
   hipStream_t s = (hipStream_t)stream;
   bool has_res = res != 0;
-#define LNB(DT, HR)                                                          \
-  hipLaunchKernelGGL((ln_bwd_kernel<DT, BLOCK, HR, RPW>), grid, dim3(BLOCK), \
-                     lds_bytes, s, (const void*)dy, (const void*)x,          \
-                     (const void*)res, (const void*)w, (const float*)mean,   \
-                     (const float*)rstd, (void*)dx, (float*)dw, (float*)db,  \
-                     rows, cols)
-  if (dt == DT_F32) { if (has_res) LNB(DT_F32, true); else LNB(DT_F32, false); }
-  else              { if (has_res) LNB(DT_BF16, true); else LNB(DT_BF16, false); }
-#undef LNB
+  if (ln_fast_ok(cols)) {
+    unsigned grid = (unsigned)((rows + 3) / 4);
+    if (grid > 8192u) grid = 8192u;
+    const int ch = (int)((cols / 8 + WAVE - 1) / WAVE);
+#define LNBV(DT, HR, CH)                                                       \
+  hipLaunchKernelGGL((ln_bwd_dx_vec_kernel<DT, HR, CH>), dim3(grid),           \
+                     dim3(256), 0, s, (const void*)dy, (const void*)x,         \
+                     (const void*)res, (const void*)w, (const float*)mean,     \
+                     (const float*)rstd, (void*)dx, rows, cols)
+#define LNBV_CH(DT, HR)                                                        \
+  do {                                                                         \
+    if (ch <= 2) LNBV(DT, HR, 2);                                              \
+    else if (ch <= 4) LNBV(DT, HR, 4);                                         \
+    else LNBV(DT, HR, 8);                                                      \
+  } while (0)
+    if (dt == DT_F32) { if (has_res) LNBV_CH(DT_F32, true); else LNBV_CH(DT_F32, false); }
+    else              { if (has_res) LNBV_CH(DT_BF16, true); else LNBV_CH(DT_BF16, false); }
+#undef LNBV_CH
+#undef LNBV
+  } else {
+    dim3 grid((unsigned)(rows < 4096 ? rows : 4096));
+#define LNBD(DT, HR)                                                           \
+  hipLaunchKernelGGL((ln_bwd_dx_kernel<DT, 256, HR>), grid, dim3(256), 0, s,   \
+                     (const void*)dy, (const void*)x, (const void*)res,        \
+                     (const void*)w, (const float*)mean, (const float*)rstd,   \
+                     (void*)dx, rows, cols)
+    if (dt == DT_F32) { if (has_res) LNBD(DT_F32, true); else LNBD(DT_F32, false); }
+    else              { if (has_res) LNBD(DT_BF16, true); else LNBD(DT_BF16, false); }
+#undef LNBD
+  }
+  {
+    constexpr int BLOCK = 256;
+    // slab sized so the grid fills 256 CUs (cols/256 col-blocks * row-slabs)
+    int64_t slab = 64;
+    while ((cols + BLOCK - 1) / BLOCK * ((rows + slab - 1) / slab) > 2048 && slab < rows)
+      slab *= 2;
+    dim3 grid((unsigned)((cols + BLOCK - 1) / BLOCK),
+              (unsigned)((rows + slab - 1) / slab));
+#define LNWB(DT, HR)                                                           \
+  hipLaunchKernelGGL((ln_bwd_wb_kernel<DT, HR, BLOCK>), grid, dim3(BLOCK), 0,  \
+                     s, (const void*)dy, (const void*)x, (const void*)res,     \
+                     (const float*)mean, (const float*)rstd, (float*)dw,       \
+                     (float*)db, rows, cols, slab)
+    if (dt == DT_F32) { if (has_res) LNWB(DT_F32, true); else LNWB(DT_F32, false); }
+    else              { if (has_res) LNWB(DT_BF16, true); else LNWB(DT_BF16, false); }
+#undef LNWB
+  }
   LAUNCH_CHECK();
   return 0;
 }

@@ -27,7 +27,30 @@ __global__ __launch_bounds__(BLOCK) void sgd_kernel(
   const void* g = (const void*)d.g;
   float* master = (float*)d.m;
   float* mbuf = (float*)d.mom;
-  for (int64_t i = threadIdx.x; i < d.n; i += BLOCK) {
+  const int64_t n8 = d.n / 8;
+  // vectorized body (slab bases are 16B-aligned; tail handled below)
+  for (int64_t i8 = threadIdx.x; i8 < n8; i8 += BLOCK) {
+    float gv[8], pv[8];
+    Vec8<DT>::load(g, i8, gv);
+    if (MASTER)
+      Vec8<DT_F32>::load(master, i8, pv);
+    else
+      Vec8<DT>::load(p, i8, pv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gg = gv[j];
+      if (wd != 0.f) gg += wd * pv[j];
+      if (MOM) {
+        float b = mbuf[i8 * 8 + j] * momentum + gg;
+        mbuf[i8 * 8 + j] = b;
+        gg = b;
+      }
+      pv[j] -= lr * gg;
+    }
+    if (MASTER) Vec8<DT_F32>::store(master, i8, pv);
+    Vec8<DT>::store(p, i8, pv);
+  }
+  for (int64_t i = n8 * 8 + threadIdx.x; i < d.n; i += BLOCK) {
     float gv = load_elem<DT>(g, i);
     float pv = MASTER ? master[i] : load_elem<DT>(p, i);
     if (wd != 0.f) gv += wd * pv;
