@@ -38,6 +38,8 @@ def parse_args():
     p.add_argument("--slowdowns", default="")
     p.add_argument("--stimulate", action="store_true")
     p.add_argument("--dropout", type=float, default=0.1)
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph step capture (1-GPU path)")
     p.add_argument("--json-out", default="")
     return p.parse_args()
 
@@ -115,8 +117,22 @@ def main():
     loader = torch.utils.data.DataLoader(ds, batch_size=args.batch, shuffle=False, drop_last=True)
     batches = list(loader)
 
+    graphed = None
+    if use_cuda and world == 1 and sd == 0 and not args.no_graph:
+        from skycomputing_amd.parallel.graph import GraphedTrainStep
+
+        data0, labels0 = batches[0]
+        graphed = GraphedTrainStep(
+            engine.stage, opt,
+            lambda logits, labels: torch.nn.functional.cross_entropy(logits.float(), labels),
+            list(data0), labels0,
+        )
+
     def step(i):
         data, labels = batches[i % len(batches)]
+        if graphed is not None:
+            graphed.step(data, labels)
+            return
         opt.zero_grad(set_to_none=True)
         engine.run_iteration(data, labels, num_microbatches=M, schedule=args.schedule)
         opt.step()
@@ -160,6 +176,7 @@ def main():
                 "allocate": args.allocate,
                 "dropout": args.dropout,
                 "num_pipeline_layers": L,
+                "hipgraph": bool(graphed is not None),
             },
         }
         line = json.dumps(result)

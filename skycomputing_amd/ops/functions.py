@@ -164,24 +164,46 @@ class MaskedSoftmaxFn(torch.autograd.Function):
         return ds, None, None
 
 
+_RNG_STATE: torch.Tensor | None = None
+
+
+def rng_state() -> torch.Tensor:
+    """Device-side step counter mixed into every dropout seed. Bump it with
+    rng_tick() at iteration start — as a captured hipGraph node this keeps
+    dropout masks varying under graph replay (launch args are frozen)."""
+    global _RNG_STATE
+    if _RNG_STATE is None:
+        _RNG_STATE = torch.zeros(1, dtype=torch.int64, device="cuda")
+    return _RNG_STATE
+
+
+def rng_tick():
+    lib = hiplib.require()
+    check(lib.sky_rng_tick(_stream(), rng_state().data_ptr()), "sky_rng_tick")
+
+
 class DropoutFn(torch.autograd.Function):
-    """Dropout with counter-based RNG: the keep mask is regenerated from the
-    seed in backward, so no mask tensor is stored."""
+    """Dropout with counter-based RNG: the keep mask is regenerated from
+    (salt, device step counter) in backward, so no mask tensor is stored
+    and the op is hipGraph-capture safe."""
 
     @staticmethod
     def forward(ctx, x, p):
         lib = hiplib.require()
         x = x.contiguous()
         y = torch.empty_like(x)
-        seed = _next_seed()
+        salt = _next_seed()
         keep = 1.0 - p
+        state = rng_state()
         check(
-            lib.sky_dropout_fwd(_stream(), ptr(x), ptr(y), x.numel(), keep, seed, _dt(x)),
+            lib.sky_dropout_fwd(
+                _stream(), ptr(x), ptr(y), x.numel(), keep, salt,
+                state.data_ptr(), _dt(x),
+            ),
             "sky_dropout_fwd",
         )
-        ctx.seed = seed
+        ctx.salt = salt
         ctx.keep = keep
-        ctx.meta = (x.numel(), x.dtype, x.device)
         return y
 
     @staticmethod
@@ -191,7 +213,8 @@ class DropoutFn(torch.autograd.Function):
         dx = torch.empty_like(dy)
         check(
             lib.sky_dropout_bwd(
-                _stream(), ptr(dy), ptr(dx), dy.numel(), ctx.keep, ctx.seed, _dt(dy)
+                _stream(), ptr(dy), ptr(dx), dy.numel(), ctx.keep, ctx.salt,
+                rng_state().data_ptr(), _dt(dy),
             ),
             "sky_dropout_bwd",
         )

@@ -162,11 +162,18 @@ SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
 // RNG indexed by ABSOLUTE element index so forward and backward agree for
 // any vectorization.
 
+// The effective RNG stream is (salt + device step counter): the counter is
+// bumped ON DEVICE by sky_rng_tick at the start of every training
+// iteration, so dropout masks vary across steps even when the launch
+// parameters are frozen inside a captured hipGraph (salt alone would be
+// constant under replay).
+
 template <int DT>
 __global__ __launch_bounds__(256) void dropout_vec_kernel(
     const void* __restrict__ x, void* __restrict__ y, int64_t n8, float keep,
-    uint64_t seed) {
+    uint64_t salt, const unsigned long long* __restrict__ state) {
   const float inv_keep = 1.f / keep;
+  const uint64_t seed = salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull;
   for (int64_t i8 = (int64_t)blockIdx.x * 256 + threadIdx.x; i8 < n8;
        i8 += (int64_t)gridDim.x * 256) {
     float v[8];
@@ -183,8 +190,9 @@ __global__ __launch_bounds__(256) void dropout_vec_kernel(
 template <int DT>
 __global__ __launch_bounds__(256) void dropout_kernel(
     const void* __restrict__ x, void* __restrict__ y, int64_t n, float keep,
-    uint64_t seed) {
+    uint64_t salt, const unsigned long long* __restrict__ state) {
   const float inv_keep = 1.f / keep;
+  const uint64_t seed = salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull;
   for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * 256) {
     float v = load_elem<DT>(x, i);
@@ -193,8 +201,20 @@ __global__ __launch_bounds__(256) void dropout_kernel(
   }
 }
 
+__global__ void rng_tick_kernel(unsigned long long* state) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) state[0] += 1;
+}
+
+SKY_EXPORT int sky_rng_tick(uint64_t stream, uint64_t state) {
+  hipLaunchKernelGGL(rng_tick_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, (unsigned long long*)state);
+  LAUNCH_CHECK();
+  return 0;
+}
+
 SKY_EXPORT int sky_dropout_fwd(uint64_t stream, uint64_t x, uint64_t y,
-                               int64_t n, float keep, uint64_t seed, int dt) {
+                               int64_t n, float keep, uint64_t salt,
+                               uint64_t state, int dt) {
   hipStream_t s = (hipStream_t)stream;
   if (n % 8 == 0) {
     int64_t n8 = n / 8;
@@ -202,26 +222,31 @@ SKY_EXPORT int sky_dropout_fwd(uint64_t stream, uint64_t x, uint64_t y,
     if (grid > 2048u) grid = 2048u;
     if (dt == DT_F32)
       hipLaunchKernelGGL((dropout_vec_kernel<DT_F32>), dim3(grid), dim3(256), 0, s,
-                         (const void*)x, (void*)y, n8, keep, seed);
+                         (const void*)x, (void*)y, n8, keep, salt,
+                         (const unsigned long long*)state);
     else
       hipLaunchKernelGGL((dropout_vec_kernel<DT_BF16>), dim3(grid), dim3(256), 0, s,
-                         (const void*)x, (void*)y, n8, keep, seed);
+                         (const void*)x, (void*)y, n8, keep, salt,
+                         (const unsigned long long*)state);
   } else {
     unsigned grid = (unsigned)((n + 255) / 256);
     if (grid > 2048u) grid = 2048u;
     if (dt == DT_F32)
       hipLaunchKernelGGL((dropout_kernel<DT_F32>), dim3(grid), dim3(256), 0, s,
-                         (const void*)x, (void*)y, n, keep, seed);
+                         (const void*)x, (void*)y, n, keep, salt,
+                         (const unsigned long long*)state);
     else
       hipLaunchKernelGGL((dropout_kernel<DT_BF16>), dim3(grid), dim3(256), 0, s,
-                         (const void*)x, (void*)y, n, keep, seed);
+                         (const void*)x, (void*)y, n, keep, salt,
+                         (const unsigned long long*)state);
   }
   LAUNCH_CHECK();
   return 0;
 }
 
 SKY_EXPORT int sky_dropout_bwd(uint64_t stream, uint64_t dy, uint64_t dx,
-                               int64_t n, float keep, uint64_t seed, int dt) {
+                               int64_t n, float keep, uint64_t salt,
+                               uint64_t state, int dt) {
   // identical math: dx = dy * mask / keep
-  return sky_dropout_fwd(stream, dy, dx, n, keep, seed, dt);
+  return sky_dropout_fwd(stream, dy, dx, n, keep, salt, state, dt);
 }

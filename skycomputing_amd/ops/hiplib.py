@@ -50,9 +50,10 @@ def _register_signatures(lib):
         "sky_sgd_step": [u64, u64, i64, f32, f32, f32, i32, i32],
         #                strm descs n   lr   mom  wd   dt   flags
         "sky_detect_mem": [u64, u64],  # free_out, total_out (host ptrs)
-        "sky_dropout_fwd": [u64, u64, u64, i64, f32, u64, i32],
-        #                   strm  x    y    n   keep seed dt
-        "sky_dropout_bwd": [u64, u64, u64, i64, f32, u64, i32],
+        "sky_dropout_fwd": [u64, u64, u64, i64, f32, u64, u64, i32],
+        #                   strm  x    y    n   keep salt state dt
+        "sky_dropout_bwd": [u64, u64, u64, i64, f32, u64, u64, i32],
+        "sky_rng_tick": [u64, u64],  # strm, state ptr
     }
     for name, argtypes in sigs.items():
         if hasattr(lib, name):

@@ -102,3 +102,65 @@ def test_graft_smoke():
     import __graft_entry__
 
     __graft_entry__.smoke()
+
+
+def test_graphed_train_step():
+    """hipGraph-captured step: finite decreasing loss, dropout mask varies
+    across replays (device RNG tick)."""
+    torch.manual_seed(2)
+    from skycomputing_amd.builder import build_module_from_cfg
+    from skycomputing_amd.models import bert_pipeline_config
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel.graph import GraphedTrainStep
+
+    cfgs = bert_pipeline_config(
+        2, dict(hidden_size=256, num_attention_heads=4, intermediate_size=1024,
+                vocab_size=2000, hidden_dropout_prob=0.1,
+                attention_probs_dropout_prob=0.1)
+    )
+    stage = build_module_from_cfg(cfgs, device="cuda:0", dtype=torch.bfloat16,
+                                  record_forward_time=False)
+    opt = FusedSGD(stage.parameters(), lr=1e-2)
+    ids = torch.randint(0, 2000, (8, 32), device="cuda")
+    inputs = [ids, torch.zeros_like(ids), torch.ones_like(ids)]
+    labels = torch.randint(0, 3, (8,), device="cuda")
+    g = GraphedTrainStep(
+        stage, opt,
+        lambda lg, lb: torch.nn.functional.cross_entropy(lg.float(), lb),
+        inputs, labels,
+    )
+    losses = []
+    for _ in range(10):
+        g.step(inputs, labels)
+        losses.append(g.loss_value())
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0], losses
+    # with dropout active, replayed losses on identical data must differ
+    # (device RNG tick) once weights stop changing much; check variation
+    assert len({round(l, 6) for l in losses}) > 1
+
+
+def test_dropout_varies_across_ticks():
+    from skycomputing_amd.ops import functions as F
+
+    x = torch.ones(4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y1 = F.DropoutFn.apply(x, 0.5)
+    F.rng_tick()
+    torch.cuda.synchronize()
+    # same salt path can't be forced (new salt per call), so check the
+    # underlying kernel directly with a fixed salt before/after tick
+    from skycomputing_amd.ops import hiplib
+    from skycomputing_amd.ops.hiplib import check, ptr
+
+    lib = hiplib.require()
+    out_a = torch.empty_like(x)
+    out_b = torch.empty_like(x)
+    st = F.rng_state()
+    stream = torch.cuda.current_stream().cuda_stream
+    check(lib.sky_dropout_fwd(stream, ptr(x), ptr(out_a), x.numel(), 0.5, 123,
+                              st.data_ptr(), 1), "f")
+    F.rng_tick()
+    check(lib.sky_dropout_fwd(stream, ptr(x), ptr(out_b), x.numel(), 0.5, 123,
+                              st.data_ptr(), 1), "f")
+    torch.cuda.synchronize()
+    assert not torch.equal(out_a, out_b)
