@@ -74,6 +74,22 @@ class LinearActivation(nn.Module):
         return ops.linear_act(x, self.weight, self.bias, self.act)
 
 
+class SkyLinear(nn.Module):
+    """nn.Linear drop-in whose backward computes dbias with the HIP column
+    reduction (ops.linear); GEMMs stay on hipBLASLt."""
+
+    def __init__(self, in_features: int, out_features: int, bias: bool = True):
+        super().__init__()
+        self.in_features = in_features
+        self.out_features = out_features
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        self.bias = nn.Parameter(torch.zeros(out_features)) if bias else None
+        nn.init.normal_(self.weight, std=0.02)
+
+    def forward(self, x):
+        return ops.linear(x, self.weight, self.bias)
+
+
 class BertSelfAttention(nn.Module):
     """Multi-head self-attention core
     (reference: scaelum/model/bert_layers.py:215-275).
@@ -89,7 +105,7 @@ class BertSelfAttention(nn.Module):
         self.num_heads = config.num_attention_heads
         self.head_dim = config.hidden_size // config.num_attention_heads
         H = config.hidden_size
-        self.qkv = nn.Linear(H, 3 * H)
+        self.qkv = SkyLinear(H, 3 * H)
         self.dropout_p = config.attention_probs_dropout_prob
 
     def forward(self, hidden, ext_mask):
@@ -111,7 +127,7 @@ class BertSelfOutput(nn.Module):
     def __init__(self, config: BertConfig):
         super().__init__()
         H = config.hidden_size
-        self.dense = nn.Linear(H, H)
+        self.dense = SkyLinear(H, H)
         self.layer_norm = SkyLayerNorm(H, config.layer_norm_eps)
         self.dropout_p = config.hidden_dropout_prob
 
@@ -235,7 +251,7 @@ class BertLayerTail(nn.Module):
         super().__init__()
         config = _cfg(config)
         self.config = config
-        self.dense = nn.Linear(config.intermediate_size, config.hidden_size)
+        self.dense = SkyLinear(config.intermediate_size, config.hidden_size)
         self.layer_norm = SkyLayerNorm(config.hidden_size, config.layer_norm_eps)
         self.dropout_p = config.hidden_dropout_prob
 

@@ -101,6 +101,15 @@ def attention_context(q, k, v, mask, dropout_p: float = 0.0, training: bool = Fa
     return eager.attention_context(q, k, v, mask, dropout_p, training)
 
 
+def linear(x, weight, bias=None):
+    """Linear with HIP-colsum dbias backward on GPU (GEMMs via hipBLASLt)."""
+    if _use_hip(x):
+        from .functions import LinearBiasFn
+
+        return LinearBiasFn.apply(x, weight, bias)
+    return torch.nn.functional.linear(x, weight, bias)
+
+
 def linear_act(x, weight, bias, act: str = "gelu"):
     """Linear + bias + activation (the reference's LinearActivation).
 

@@ -221,6 +221,44 @@ class DropoutFn(torch.autograd.Function):
         return dx, None
 
 
+def colsum(src: torch.Tensor, out_dtype=None) -> torch.Tensor:
+    """Column sum over a 2D [rows, cols] tensor via the HIP column-parallel
+    reduction (fp32 accumulation)."""
+    lib = hiplib.require()
+    src = src.contiguous()
+    rows, cols = src.shape
+    out = torch.zeros(cols, dtype=torch.float32, device=src.device)
+    check(lib.sky_colsum(_stream(), ptr(src), ptr(out), rows, cols, _dt(src)), "sky_colsum")
+    return out.to(out_dtype) if out_dtype is not None else out
+
+
+class LinearBiasFn(torch.autograd.Function):
+    """Linear + bias with a custom backward: dgrad/wgrad stay hipBLASLt
+    GEMMs, but dbias uses the HIP column reduction instead of torch's
+    generic reduce_kernel (a measured hot spot — profiles/r01_notes.md)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1])
+        y = torch.nn.functional.linear(x2, weight, bias)
+        ctx.save_for_backward(x2, weight)
+        ctx.xshape = xs
+        ctx.has_bias = bias is not None
+        return y.view(*xs[:-1], weight.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, weight = ctx.saved_tensors
+        dy2 = dy.reshape(-1, weight.shape[0])
+        if not dy2.is_contiguous():
+            dy2 = dy2.contiguous()
+        dx = dy2.mm(weight).view(ctx.xshape)
+        dw = dy2.t().mm(x2)
+        db = colsum(dy2, weight.dtype) if ctx.has_bias else None
+        return dx, dw, db
+
+
 class EmbeddingFusedFn(torch.autograd.Function):
     """Fused word+position+type gather, 3-way add and LayerNorm
     (reference eager sequence: scaelum/model/bert_layers.py:191-212)."""

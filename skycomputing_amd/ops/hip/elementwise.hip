@@ -125,6 +125,17 @@ static void launch_colsum(hipStream_t s, const void* src, float* out,
                      out, rows, cols, slab);
 }
 
+SKY_EXPORT int sky_colsum(uint64_t stream, uint64_t src, uint64_t out,
+                          int64_t rows, int64_t cols, int dt) {
+  hipStream_t s = (hipStream_t)stream;
+  if (dt == DT_F32)
+    launch_colsum<DT_F32>(s, (const void*)src, (float*)out, rows, cols);
+  else
+    launch_colsum<DT_BF16>(s, (const void*)src, (float*)out, rows, cols);
+  LAUNCH_CHECK();
+  return 0;
+}
+
 SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
                                  uint64_t b, uint64_t dx, uint64_t db,
                                  int64_t rows, int64_t cols, int dt) {

@@ -54,6 +54,7 @@ def _register_signatures(lib):
         #                   strm  x    y    n   keep salt state dt
         "sky_dropout_bwd": [u64, u64, u64, i64, f32, u64, u64, i32],
         "sky_rng_tick": [u64, u64],  # strm, state ptr
+        "sky_colsum": [u64, u64, u64, i64, i64, i32],  # strm src out32 rows cols dt
     }
     for name, argtypes in sigs.items():
         if hasattr(lib, name):
