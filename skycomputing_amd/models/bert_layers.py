@@ -110,14 +110,8 @@ class BertSelfAttention(nn.Module):
 
     def forward(self, hidden, ext_mask):
         B, S, H = hidden.shape
-        h, d = self.num_heads, self.head_dim
-        qkv = self.qkv(hidden).view(B, S, 3, h, d)
-        # strided views; batched GEMMs consume them without materializing
-        q = qkv[:, :, 0].permute(0, 2, 1, 3)
-        k = qkv[:, :, 1].permute(0, 2, 1, 3)
-        v = qkv[:, :, 2].permute(0, 2, 1, 3)
-        ctx = ops.attention_context(q, k, v, ext_mask, self.dropout_p, self.training)
-        return ctx.permute(0, 2, 1, 3).reshape(B, S, h * d)
+        qkv = self.qkv(hidden).view(B, S, 3, self.num_heads, self.head_dim)
+        return ops.attention(qkv, ext_mask, self.dropout_p, self.training)
 
 
 class BertSelfOutput(nn.Module):

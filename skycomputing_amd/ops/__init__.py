@@ -101,6 +101,30 @@ def attention_context(q, k, v, mask, dropout_p: float = 0.0, training: bool = Fa
     return eager.attention_context(q, k, v, mask, dropout_p, training)
 
 
+def attention(qkv, mask, dropout_p: float = 0.0, training: bool = False):
+    """Multi-head attention over a packed qkv tensor [B, S, 3, h, d];
+    returns [B, S, h*d].
+
+    GPU fast path (bf16, d=64, S<=128): ONE fused MFMA kernel
+    (FusedAttentionFn). Otherwise: strided-view batched GEMMs + fused
+    masked softmax (HIP on GPU, eager on CPU)."""
+    B, S, three, h, d = qkv.shape
+    scale = 1.0 / math.sqrt(d)
+    if (
+        qkv.is_cuda and qkv.dtype == torch.bfloat16 and d == 64 and S <= 128
+        and hiplib.available() and os.environ.get("SKY_NO_FUSED_ATTN") != "1"
+    ):
+        from .functions import FusedAttentionFn
+
+        out = FusedAttentionFn.apply(qkv, mask, scale, dropout_p, training)
+        return out.reshape(B, S, h * d)
+    q = qkv[:, :, 0].permute(0, 2, 1, 3)
+    k = qkv[:, :, 1].permute(0, 2, 1, 3)
+    v = qkv[:, :, 2].permute(0, 2, 1, 3)
+    ctx = attention_context(q, k, v, mask, dropout_p, training)
+    return ctx.permute(0, 2, 1, 3).reshape(B, S, h * d)
+
+
 def linear(x, weight, bias=None):
     """Linear with HIP-colsum dbias backward on GPU (GEMMs via hipBLASLt)."""
     if _use_hip(x):

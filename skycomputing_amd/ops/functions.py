@@ -221,6 +221,91 @@ class DropoutFn(torch.autograd.Function):
         return dx, None
 
 
+class FusedAttentionFn(torch.autograd.Function):
+    """Fused attention: out = dropout(softmax(scale*QK^T + mask)) V in ONE
+    MFMA kernel (ops/hip/attention.hip); S x S probabilities are never
+    materialized in forward (row max/sum saved, flash-style). Backward
+    recomputes P (+ dropout mask) with sky_attn_probs, then runs
+    dV/dP/dS/dQ/dK as hipBLASLt batched GEMMs + the HIP softmax-backward.
+
+    qkv: [B, S, 3, h, d] (d = 64), mask: [B, 1, 1, S] additive or None.
+    Returns [B, S, h, d].
+    """
+
+    @staticmethod
+    def forward(ctx, qkv, mask, scale, dropout_p, training):
+        lib = hiplib.require()
+        B, S, three, h, d = qkv.shape
+        assert three == 3 and d == 64 and S <= 128
+        qkv = qkv.contiguous()
+        mask = mask.contiguous() if mask is not None else None
+        out = torch.empty(B, S, h, d, dtype=qkv.dtype, device=qkv.device)
+        m = torch.empty(B, h, S, dtype=torch.float32, device=qkv.device)
+        lsum = torch.empty_like(m)
+        keep = 1.0 - dropout_p if (dropout_p > 0 and training) else 1.0
+        salt = _next_seed() if keep < 1.0 else 0
+        state = rng_state()
+        check(
+            lib.sky_attn_fwd(
+                _stream(), ptr(qkv), ptr(mask), ptr(out), ptr(m), ptr(lsum),
+                B, S, h, d, scale, keep, salt, state.data_ptr(),
+            ),
+            "sky_attn_fwd",
+        )
+        ctx.save_for_backward(qkv, mask, m, lsum)
+        ctx.scale, ctx.keep, ctx.salt = scale, keep, salt
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        lib = hiplib.require()
+        qkv, mask, m, lsum = ctx.saved_tensors
+        B, S, _, h, d = qkv.shape
+        dev = qkv.device
+        P = torch.empty(B, h, S, S, dtype=qkv.dtype, device=dev)
+        Pd = torch.empty_like(P) if ctx.keep < 1.0 else P
+        check(
+            lib.sky_attn_probs(
+                _stream(), ptr(qkv), ptr(mask), ptr(m), ptr(lsum), ptr(P),
+                ptr(Pd), B, S, h, d, ctx.scale, ctx.keep, ctx.salt,
+                rng_state().data_ptr(),
+            ),
+            "sky_attn_probs",
+        )
+        dO = dout.contiguous().permute(0, 2, 1, 3)  # [B,h,S,d] strided view
+        q = qkv[:, :, 0].permute(0, 2, 1, 3)
+        k = qkv[:, :, 1].permute(0, 2, 1, 3)
+        v = qkv[:, :, 2].permute(0, 2, 1, 3)
+        dV = torch.matmul(Pd.transpose(-1, -2), dO)
+        dPd = torch.matmul(dO, v.transpose(-1, -2)).contiguous()
+        if ctx.keep < 1.0:
+            dP = torch.empty_like(dPd)
+            check(
+                lib.sky_dropout_bwd(
+                    _stream(), ptr(dPd), ptr(dP), dPd.numel(), ctx.keep,
+                    ctx.salt, rng_state().data_ptr(), _dt(dPd),
+                ),
+                "sky_dropout_bwd",
+            )
+        else:
+            dP = dPd
+        dS = torch.empty_like(dP)
+        check(
+            lib.sky_masked_softmax_bwd(
+                _stream(), ptr(dP), ptr(P), ptr(dS), B, h, S, S, ctx.scale,
+                1.0, 0, _dt(P),
+            ),
+            "sky_masked_softmax_bwd",
+        )
+        dQ = torch.matmul(dS, k)                      # [B,h,S,d]
+        dK = torch.matmul(dS.transpose(-1, -2), q)    # [B,h,S,d]
+        dqkv = torch.empty_like(qkv)
+        dqkv[:, :, 0] = dQ.permute(0, 2, 1, 3)
+        dqkv[:, :, 1] = dK.permute(0, 2, 1, 3)
+        dqkv[:, :, 2] = dV.permute(0, 2, 1, 3)
+        return dqkv, None, None, None, None
+
+
 def colsum(src: torch.Tensor, out_dtype=None) -> torch.Tensor:
     """Column sum over a 2D [rows, cols] tensor via the HIP column-parallel
     reduction (fp32 accumulation)."""

@@ -1,0 +1,318 @@
+// Fused multi-head attention forward for gfx950 (MFMA 16x16x32 bf16).
+//
+// Computes out = dropout(softmax(scale * Q K^T + mask)) V for one (batch,
+// head) per workgroup, S <= 128, d = 64 (BERT-large heads) — replacing the
+// reference's QK^T GEMM + mask-add + softmax + dropout + PV GEMM + head
+// permutes (reference: scaelum/model/bert_layers.py:249-275) with ONE
+// kernel. The S x S score matrix never touches HBM; row max / row sum are
+// saved (fp32) so backward recomputes probabilities with sky_attn_probs.
+//
+// Layouts (chosen so no transpose copies happen anywhere in the layer):
+//   qkv  [B, S, 3, h, d]  — straight out of the fused QKV GEMM
+//   out  [B, S, h, d]     — views as [B, S, h*d] for the output projection
+//
+// Structure per workgroup (4 waves, 256 threads):
+//   * K tile staged to LDS [S][64] (XOR-swizzled), V staged TRANSPOSED to
+//     LDS [64][S] so PV's B-fragments are contiguous ds_read_b128;
+//   * each wave owns 32 query rows: QK^T via mfma_f32_16x16x32_bf16
+//     (A = Q fragments straight from global, B = K from LDS), softmax in
+//     registers (subwave shuffles), dropout via the counter RNG, P written
+//     to a per-wave LDS tile, PV via MFMA (A = P, B = V^T from LDS).
+//
+// Fragment mappings verified on hardware by sky_mfma_probe
+// (tests/test_ops_gpu.py::test_mfma_layout).
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define ATT_SMAX 128
+#define ATT_D 64
+
+// LDS layout (bytes): K [128][64]bf16 at 0 (16K), Vt [64][128]bf16 at 16K,
+// P per wave [32][128]bf16 at 32K + w*8K. Total 64K.
+#define K_OFF 0
+#define VT_OFF (16 * 1024)
+#define P_OFF (32 * 1024)
+
+DEV void* lds_at(char* base, int byte) { return (void*)(base + byte); }
+
+DEV int swz(int byte_addr, int row, int rmask) {
+  return byte_addr ^ ((row & rmask) << 4);
+}
+
+template <bool HAS_MASK, bool SAVE_ML, bool PROBS_MODE>
+__global__ __launch_bounds__(256) void attn_fwd_kernel(
+    const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ mask,
+    ushort_t* __restrict__ out, float* __restrict__ m_io,
+    float* __restrict__ l_io, ushort_t* __restrict__ p_out,
+    ushort_t* __restrict__ pd_out, int B, int S, int h, float scale,
+    float keep, uint64_t salt, const unsigned long long* __restrict__ state) {
+  extern __shared__ __attribute__((aligned(16))) char lds[];
+  const int tid = threadIdx.x;
+  const int l = tid & 63;
+  const int w = tid >> 6;
+  const int bh = blockIdx.x;
+  const int b = bh / h;
+  const int hh = bh % h;
+  const int ts = 3 * h * ATT_D;  // token stride in qkv
+  const int NT = (S + 15) / 16;  // 16-token k tiles
+  const uint64_t seed =
+      salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull;
+  const float inv_keep = 1.f / keep;
+
+  const ushort_t* qbase = qkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
+  const ushort_t* kbase = qbase + (size_t)h * ATT_D;
+  const ushort_t* vbase = qbase + (size_t)2 * h * ATT_D;
+
+  // ---- zero LDS when S < full tile coverage (avoid NaN poisoning) ----
+  if (S < ATT_SMAX) {
+    for (int u = tid; u < (64 * 1024) / 16; u += 256)
+      *(ushort8_t*)lds_at(lds, u * 16) = (ushort8_t)(ushort_t)0;
+    __syncthreads();
+  } else {
+    // P tiles still need zeroing for partial last q tile? S==128 -> full.
+  }
+
+  // ---- stage K into LDS [S][64] (swizzled rows of 128 B) ----
+  for (int u = tid; u < S * 8; u += 256) {
+    const int tok = u >> 3;
+    const int c16 = u & 7;
+    ushort8_t v = *(const ushort8_t*)(kbase + (size_t)tok * ts + c16 * 8);
+    *(ushort8_t*)lds_at(lds, swz(K_OFF + tok * 128 + c16 * 16, tok, 7)) = v;
+  }
+  // ---- stage V transposed into LDS [64][S] ----
+  for (int u = tid; u < S * 8; u += 256) {
+    const int tok = u >> 3;
+    const int c16 = u & 7;
+    ushort8_t v = *(const ushort8_t*)(vbase + (size_t)tok * ts + c16 * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c16 * 8 + j;
+      *(ushort_t*)lds_at(lds, swz(VT_OFF + c * 256 + tok * 2, c, 15)) = v[j];
+    }
+  }
+  __syncthreads();
+
+  // ---- per-wave: 32 q rows = q tiles {2w, 2w+1} ----
+  const int qt0 = 2 * w;
+  const int lm = l & 15;
+  const int lg = l >> 4;  // lane group 0..3
+
+  // additive mask values for this lane's k columns (col = kt*16 + lm)
+  float mval[8];
+#pragma unroll
+  for (int kt = 0; kt < 8; ++kt) {
+    const int col = kt * 16 + lm;
+    float mv = 0.f;
+    if (HAS_MASK && col < S) mv = bf16_to_f32(mask[(size_t)b * S + col]);
+    mval[kt] = (col < S) ? mv : -3.0e38f;
+  }
+
+  f32x4 oacc[2][4];
+#pragma unroll
+  for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+    for (int dv = 0; dv < 4; ++dv) oacc[qi][dv] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  for (int qi = 0; qi < 2; ++qi) {
+    const int qtile = qt0 + qi;
+    const int qtok_base = qtile * 16;
+    if (qtok_base >= S) break;
+    // A fragments of Q (two K=32 splits of d=64)
+    bf16x8 aq[2];
+    {
+      int qtok = qtok_base + lm;
+      if (qtok >= S) qtok = S - 1;  // clamped read; row skipped at write
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        aq[ks] = *(const bf16x8*)(qbase + (size_t)qtok * ts + ks * 32 + lg * 8);
+    }
+    f32x4 sacc[8];
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) sacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      if (kt >= NT) continue;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int ktok = kt * 16 + lm;
+        bf16x8 bk = *(const bf16x8*)lds_at(
+            lds, swz(K_OFF + ktok * 128 + (ks * 32 + lg * 8) * 2, ktok, 7));
+        sacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, sacc[kt], 0, 0, 0);
+      }
+    }
+    // softmax over this wave's 4 rows per (lg, r)
+    float z[8][4];
+    float mx[4], sm[4], inv[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) mx[r] = -3.0e38f;
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      if (kt >= NT) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float zz = sacc[kt][r] * scale + mval[kt];
+        z[kt][r] = zz;
+        mx[r] = fmaxf(mx[r], zz);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        mx[r] = fmaxf(mx[r], __shfl_xor(mx[r], off, 64));
+    }
+    if (SAVE_ML) {
+      // nothing: m loaded below in PROBS_MODE only
+    }
+    float mrow[4], lrow[4];
+    if (PROBS_MODE) {
+      // use saved statistics for exact recomputation
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qtok_base + lg * 4 + r;
+        const int rr = row < S ? row : S - 1;
+        mrow[r] = m_io[((size_t)bh) * S + rr];
+        lrow[r] = l_io[((size_t)bh) * S + rr];
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) sm[r] = 0.f;
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      if (kt >= NT) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float p = __expf(z[kt][r] - (PROBS_MODE ? mrow[r] : mx[r]));
+        z[kt][r] = p;
+        sm[r] += p;
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) sm[r] += __shfl_xor(sm[r], off, 64);
+      inv[r] = 1.f / (PROBS_MODE ? lrow[r] : sm[r]);
+    }
+    if (SAVE_ML && lm == 0) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qtok_base + lg * 4 + r;
+        if (row < S) {
+          m_io[((size_t)bh) * S + row] = mx[r];
+          l_io[((size_t)bh) * S + row] = sm[r];
+        }
+      }
+    }
+    // normalize, dropout, write P (LDS in fwd mode; global in probs mode)
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      if (kt >= NT) continue;
+      const int col = kt * 16 + lm;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qtok_base + lg * 4 + r;
+        float p = z[kt][r] * inv[r];
+        float pd = p;
+        if (keep < 1.f) {
+          const uint64_t idx = ((uint64_t)bh * S + row) * S + col;
+          pd = (rng_uniform(seed, idx) < keep) ? p * inv_keep : 0.f;
+        }
+        if (PROBS_MODE) {
+          if (row < S && col < S) {
+            p_out[((size_t)bh * S + row) * S + col] = f32_to_bf16(p);
+            pd_out[((size_t)bh * S + row) * S + col] = f32_to_bf16(pd);
+          }
+        } else {
+          const int rl = qi * 16 + lg * 4 + r;  // local row 0..31
+          *(ushort_t*)lds_at(
+              lds, swz(P_OFF + w * 8192 + rl * 256 + col * 2, rl, 15)) =
+              f32_to_bf16((row < S && col < S) ? pd : 0.f);
+        }
+      }
+    }
+  }
+
+  if (PROBS_MODE) return;
+
+  // ---- PV: O[32 rows][64] per wave ----
+  // (same-wave LDS write->read; compiler inserts the lgkm waits)
+  for (int qi = 0; qi < 2; ++qi) {
+    if ((qt0 + qi) * 16 >= S) break;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      if (ks * 32 >= S) break;
+      // A fragment: P rows
+      const int rl = qi * 16 + lm;
+      bf16x8 ap = *(const bf16x8*)lds_at(
+          lds, swz(P_OFF + w * 8192 + rl * 256 + (ks * 32 + lg * 8) * 2, rl, 15));
+#pragma unroll
+      for (int dvt = 0; dvt < 4; ++dvt) {
+        const int dv = dvt * 16 + lm;
+        bf16x8 bv = *(const bf16x8*)lds_at(
+            lds, swz(VT_OFF + dv * 256 + (ks * 32 + lg * 8) * 2, dv, 15));
+        oacc[qi][dvt] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, oacc[qi][dvt], 0, 0, 0);
+      }
+    }
+  }
+  // write out [B,S,h,d]
+#pragma unroll
+  for (int qi = 0; qi < 2; ++qi) {
+#pragma unroll
+    for (int dvt = 0; dvt < 4; ++dvt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = (qt0 + qi) * 16 + lg * 4 + r;
+        const int dv = dvt * 16 + lm;
+        if (row < S)
+          out[(((size_t)b * S + row) * h + hh) * ATT_D + dv] =
+              f32_to_bf16(oacc[qi][dvt][r]);
+      }
+    }
+  }
+}
+
+static int attn_launch(uint64_t stream, uint64_t qkv, uint64_t mask,
+                       uint64_t out, uint64_t m, uint64_t lsum, uint64_t p,
+                       uint64_t pd, int64_t B, int64_t S, int64_t h,
+                       int64_t d, float scale, float keep, uint64_t salt,
+                       uint64_t state, bool probs_mode) {
+  if (d != ATT_D || S > ATT_SMAX) return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid((unsigned)(B * h));
+  size_t lds_bytes = 64 * 1024;
+  bool hm = mask != 0;
+#define ATT(HM, SML, PM)                                                      \
+  hipLaunchKernelGGL((attn_fwd_kernel<HM, SML, PM>), grid, dim3(256),         \
+                     lds_bytes, s, (const ushort_t*)qkv,                      \
+                     (const ushort_t*)mask, (ushort_t*)out, (float*)m,        \
+                     (float*)lsum, (ushort_t*)p, (ushort_t*)pd, (int)B,       \
+                     (int)S, (int)h, scale, keep, salt,                       \
+                     (const unsigned long long*)state)
+  if (probs_mode) { if (hm) ATT(true, false, true); else ATT(false, false, true); }
+  else            { if (hm) ATT(true, true, false); else ATT(false, true, false); }
+#undef ATT
+  LAUNCH_CHECK();
+  return 0;
+}
+
+SKY_EXPORT int sky_attn_fwd(uint64_t stream, uint64_t qkv, uint64_t mask,
+                            uint64_t out, uint64_t m, uint64_t lsum,
+                            int64_t B, int64_t S, int64_t h, int64_t d,
+                            float scale, float keep, uint64_t salt,
+                            uint64_t state) {
+  return attn_launch(stream, qkv, mask, out, m, lsum, 0, 0, B, S, h, d,
+                     scale, keep, salt, state, false);
+}
+
+SKY_EXPORT int sky_attn_probs(uint64_t stream, uint64_t qkv, uint64_t mask,
+                              uint64_t m, uint64_t lsum, uint64_t p,
+                              uint64_t pd, int64_t B, int64_t S, int64_t h,
+                              int64_t d, float scale, float keep,
+                              uint64_t salt, uint64_t state) {
+  return attn_launch(stream, qkv, mask, 0, m, lsum, p, pd, B, S, h, d, scale,
+                     keep, salt, state, true);
+}

@@ -55,6 +55,11 @@ def _register_signatures(lib):
         "sky_dropout_bwd": [u64, u64, u64, i64, f32, u64, u64, i32],
         "sky_rng_tick": [u64, u64],  # strm, state ptr
         "sky_colsum": [u64, u64, u64, i64, i64, i32],  # strm src out32 rows cols dt
+        "sky_mfma_probe": [u64, u64, u64, u64],  # strm A B D
+        "sky_attn_fwd": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, u64],
+        #                strm qkv  mask out  m    l    B    S    h    d  scale keep salt state
+        "sky_attn_probs": [u64, u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, u64],
+        #                  strm qkv mask m    l    p    pd   B    S    h    d  scale keep salt state
     }
     for name, argtypes in sigs.items():
         if hasattr(lib, name):
