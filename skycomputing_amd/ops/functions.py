@@ -272,7 +272,7 @@ class FusedAttentionFn(torch.autograd.Function):
             ),
             "sky_attn_probs",
         )
-        dO = dout.contiguous().permute(0, 2, 1, 3)  # [B,h,S,d] strided view
+        dO = dout.permute(0, 2, 1, 3).contiguous()  # one copy, used twice
         q = qkv[:, :, 0].permute(0, 2, 1, 3)
         k = qkv[:, :, 1].permute(0, 2, 1, 3)
         v = qkv[:, :, 2].permute(0, 2, 1, 3)
@@ -300,9 +300,13 @@ class FusedAttentionFn(torch.autograd.Function):
         dQ = torch.matmul(dS, k)                      # [B,h,S,d]
         dK = torch.matmul(dS.transpose(-1, -2), q)    # [B,h,S,d]
         dqkv = torch.empty_like(qkv)
-        dqkv[:, :, 0] = dQ.permute(0, 2, 1, 3)
-        dqkv[:, :, 1] = dK.permute(0, 2, 1, 3)
-        dqkv[:, :, 2] = dV.permute(0, 2, 1, 3)
+        check(
+            lib.sky_pack3(
+                _stream(), ptr(dQ.contiguous()), ptr(dK.contiguous()),
+                ptr(dV.contiguous()), ptr(dqkv), B, S, h, d, _dt(qkv),
+            ),
+            "sky_pack3",
+        )
         return dqkv, None, None, None, None
 
 

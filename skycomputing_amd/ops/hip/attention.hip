@@ -275,6 +275,53 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   }
 }
 
+// pack dQ/dK/dV (each [B,h,S,d] contiguous) into dqkv [B,S,3,h,d] in one
+// pass — replaces three permute-copies in the attention backward.
+template <int DT>
+__global__ __launch_bounds__(256) void pack3_kernel(
+    const void* __restrict__ s0, const void* __restrict__ s1,
+    const void* __restrict__ s2, void* __restrict__ dst, int64_t B,
+    int64_t S, int64_t h, int64_t d8) {
+  const int64_t n = B * h * S * d8;
+  for (int64_t u = (int64_t)blockIdx.x * 256 + threadIdx.x; u < n;
+       u += (int64_t)gridDim.x * 256) {
+    int64_t t = u;
+    const int64_t di = t % d8; t /= d8;
+    const int64_t tok = t % S; t /= S;
+    const int64_t hh = t % h;
+    const int64_t b = t / h;
+    const int64_t src_off = u;  // [B,h,S,d] contiguous
+    const int64_t dst_base = (((b * S + tok) * 3) * h + hh) * d8 + di;
+    float v[8];
+    Vec8<DT>::load(s0, src_off, v);
+    Vec8<DT>::store(dst, dst_base, v);
+    Vec8<DT>::load(s1, src_off, v);
+    Vec8<DT>::store(dst, dst_base + h * d8, v);
+    Vec8<DT>::load(s2, src_off, v);
+    Vec8<DT>::store(dst, dst_base + 2 * h * d8, v);
+  }
+}
+
+SKY_EXPORT int sky_pack3(uint64_t stream, uint64_t s0, uint64_t s1,
+                         uint64_t s2, uint64_t dst, int64_t B, int64_t S,
+                         int64_t h, int64_t d, int dt) {
+  if (d % 8 != 0) return (int)hipErrorInvalidValue;
+  const int64_t n = B * h * S * (d / 8);
+  unsigned grid = (unsigned)((n + 255) / 256);
+  if (grid > 2048u) grid = 2048u;
+  hipStream_t s = (hipStream_t)stream;
+  if (dt == DT_F32)
+    hipLaunchKernelGGL((pack3_kernel<DT_F32>), dim3(grid), dim3(256), 0, s,
+                       (const void*)s0, (const void*)s1, (const void*)s2,
+                       (void*)dst, B, S, h, d / 8);
+  else
+    hipLaunchKernelGGL((pack3_kernel<DT_BF16>), dim3(grid), dim3(256), 0, s,
+                       (const void*)s0, (const void*)s1, (const void*)s2,
+                       (void*)dst, B, S, h, d / 8);
+  LAUNCH_CHECK();
+  return 0;
+}
+
 static int attn_launch(uint64_t stream, uint64_t qkv, uint64_t mask,
                        uint64_t out, uint64_t m, uint64_t lsum, uint64_t p,
                        uint64_t pd, int64_t B, int64_t S, int64_t h,

@@ -56,6 +56,7 @@ def _register_signatures(lib):
         "sky_rng_tick": [u64, u64],  # strm, state ptr
         "sky_colsum": [u64, u64, u64, i64, i64, i32],  # strm src out32 rows cols dt
         "sky_mfma_probe": [u64, u64, u64, u64],  # strm A B D
+        "sky_pack3": [u64, u64, u64, u64, u64, i64, i64, i64, i64, i32],
         "sky_attn_fwd": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, u64],
         #                strm qkv  mask out  m    l    B    S    h    d  scale keep salt state
         "sky_attn_probs": [u64, u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, u64],
