@@ -31,6 +31,15 @@ def _stream() -> int:
     return torch.cuda.current_stream().cuda_stream
 
 
+_CS_SLABS = 128
+
+
+def _red_scratch(cols: int, pairs: int, device) -> torch.Tensor:
+    """fp32 scratch for the two-stage column reductions:
+    [128 slabs][pairs*cols]."""
+    return torch.empty(_CS_SLABS * pairs * cols, dtype=torch.float32, device=device)
+
+
 def _dt(t: torch.Tensor) -> int:
     try:
         return _DT[t.dtype]
@@ -73,13 +82,16 @@ class LayerNormFn(torch.autograd.Function):
         cols = x.shape[-1]
         rows = x.numel() // cols
         dx = torch.empty_like(x)
-        dw32 = torch.zeros(cols, dtype=torch.float32, device=x.device)
-        db32 = torch.zeros(cols, dtype=torch.float32, device=x.device)
+        fast = cols % 8 == 0
+        alloc = torch.empty if fast else torch.zeros
+        dw32 = alloc(cols, dtype=torch.float32, device=x.device)
+        db32 = alloc(cols, dtype=torch.float32, device=x.device)
+        scratch = _red_scratch(cols, 2, x.device) if fast else None
         check(
             lib.sky_layernorm_bwd(
                 _stream(), ptr(dy), ptr(x), ptr(ctx.residual), ptr(weight),
                 ptr(mean), ptr(rstd), ptr(dx), ptr(dw32), ptr(db32),
-                rows, cols, _dt(x),
+                ptr(scratch), rows, cols, _dt(x),
             ),
             "sky_layernorm_bwd",
         )
@@ -113,10 +125,13 @@ class BiasGeluFn(torch.autograd.Function):
         cols = x.shape[-1]
         rows = x.numel() // cols
         dx = torch.empty_like(x)
-        db32 = torch.zeros(cols, dtype=torch.float32, device=x.device)
+        fast = cols % 8 == 0
+        db32 = (torch.empty if fast else torch.zeros)(cols, dtype=torch.float32, device=x.device)
+        scratch = _red_scratch(cols, 1, x.device) if fast else None
         check(
             lib.sky_bias_gelu_bwd(
-                _stream(), ptr(dy), ptr(x), ptr(bias), ptr(dx), ptr(db32), rows, cols, _dt(x)
+                _stream(), ptr(dy), ptr(x), ptr(bias), ptr(dx), ptr(db32),
+                ptr(scratch), rows, cols, _dt(x)
             ),
             "sky_bias_gelu_bwd",
         )
@@ -316,8 +331,13 @@ def colsum(src: torch.Tensor, out_dtype=None) -> torch.Tensor:
     lib = hiplib.require()
     src = src.contiguous()
     rows, cols = src.shape
-    out = torch.zeros(cols, dtype=torch.float32, device=src.device)
-    check(lib.sky_colsum(_stream(), ptr(src), ptr(out), rows, cols, _dt(src)), "sky_colsum")
+    fast = cols % 8 == 0
+    out = (torch.empty if fast else torch.zeros)(cols, dtype=torch.float32, device=src.device)
+    scratch = _red_scratch(cols, 1, src.device) if fast else None
+    check(
+        lib.sky_colsum(_stream(), ptr(src), ptr(out), ptr(scratch), rows, cols, _dt(src)),
+        "sky_colsum",
+    )
     return out.to(out_dtype) if out_dtype is not None else out
 
 

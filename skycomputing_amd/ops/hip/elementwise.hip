@@ -98,16 +98,20 @@ __global__ __launch_bounds__(256) void bias_gelu_bwd_dx_kernel(
   }
 }
 
-// db[c] = sum over rows of dx[r, c]. Vectorized: thread t owns 8
-// consecutive columns (one 16 B load per row), walking a row slab down;
-// one atomicAdd per owned column per slab.
+// db[c] = sum over rows of dx[r, c] — two-stage, no atomics, outputs need
+// no zero-init: stage 1 writes per-slab partials [nslabs][cols] fp32 to a
+// scratch buffer (thread owns 8 consecutive columns, 16 B loads); stage 2
+// reduces the slab axis and WRITES the result.
+#define CS_SLABS 128
+
 template <int DT, int BLOCK>
-__global__ __launch_bounds__(BLOCK) void colsum_vec_kernel(
-    const void* __restrict__ src, float* __restrict__ out, int64_t rows,
+__global__ __launch_bounds__(BLOCK) void colsum_part_kernel(
+    const void* __restrict__ src, float* __restrict__ scratch, int64_t rows,
     int64_t cols8, int64_t rows_per_slab) {
   const int64_t c8 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
   if (c8 >= cols8) return;
-  const int64_t r0 = (int64_t)blockIdx.y * rows_per_slab;
+  const int64_t slab = blockIdx.y;
+  const int64_t r0 = slab * rows_per_slab;
   const int64_t r1 = min(rows, r0 + rows_per_slab);
   float s[8] = {0.f};
   for (int64_t r = r0; r < r1; ++r) {
@@ -117,7 +121,18 @@ __global__ __launch_bounds__(BLOCK) void colsum_vec_kernel(
     for (int j = 0; j < 8; ++j) s[j] += v[j];
   }
 #pragma unroll
-  for (int j = 0; j < 8; ++j) atomicAdd(&out[c8 * 8 + j], s[j]);
+  for (int j = 0; j < 8; ++j) scratch[(slab * cols8 + c8) * 8 + j] = s[j];
+}
+
+template <int BLOCK>
+__global__ __launch_bounds__(BLOCK) void colsum_final_kernel(
+    const float* __restrict__ scratch, float* __restrict__ out, int64_t cols,
+    int64_t nslabs) {
+  const int64_t col = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  if (col >= cols) return;
+  float s = 0.f;
+  for (int64_t y = 0; y < nslabs; ++y) s += scratch[y * cols + col];
+  out[col] = s;
 }
 
 template <int DT, int BLOCK>
@@ -133,20 +148,26 @@ __global__ __launch_bounds__(BLOCK) void colsum_kernel(
   atomicAdd(&out[col], s);
 }
 
+// scratch must hold CS_SLABS * cols floats when the fast path is taken
+// (cols % 8 == 0); pass scratch = nullptr to force the atomic fallback
+// (out must then be pre-zeroed).
 template <int DT>
 static void launch_colsum(hipStream_t s, const void* src, float* out,
-                          int64_t rows, int64_t cols) {
-  constexpr int BLOCK = 256;
-  if (cols % 8 == 0) {
+                          float* scratch, int64_t rows, int64_t cols) {
+  if (cols % 8 == 0 && scratch != nullptr) {
+    constexpr int BLOCK = 128;
     const int64_t cols8 = cols / 8;
-    const int64_t cb = (cols8 + BLOCK - 1) / BLOCK;
-    int64_t slab = 16;
-    while (cb * ((rows + slab - 1) / slab) > 1024 && slab < rows) slab *= 2;
-    dim3 grid((unsigned)cb, (unsigned)((rows + slab - 1) / slab));
-    hipLaunchKernelGGL((colsum_vec_kernel<DT, BLOCK>), grid, dim3(BLOCK), 0,
-                       s, src, out, rows, cols8, slab);
+    const int64_t nslabs = rows < CS_SLABS ? rows : CS_SLABS;
+    const int64_t slab = (rows + nslabs - 1) / nslabs;
+    dim3 grid((unsigned)((cols8 + BLOCK - 1) / BLOCK), (unsigned)nslabs);
+    hipLaunchKernelGGL((colsum_part_kernel<DT, BLOCK>), grid, dim3(BLOCK), 0,
+                       s, src, scratch, rows, cols8, slab);
+    dim3 g2((unsigned)((cols + 255) / 256));
+    hipLaunchKernelGGL((colsum_final_kernel<256>), g2, dim3(256), 0, s,
+                       scratch, out, cols, nslabs);
     return;
   }
+  constexpr int BLOCK = 256;
   int64_t slab = 64;
   while ((cols + BLOCK - 1) / BLOCK * ((rows + slab - 1) / slab) > 2048 && slab < rows)
     slab *= 2;
@@ -157,19 +178,21 @@ static void launch_colsum(hipStream_t s, const void* src, float* out,
 }
 
 SKY_EXPORT int sky_colsum(uint64_t stream, uint64_t src, uint64_t out,
-                          int64_t rows, int64_t cols, int dt) {
+                          uint64_t scratch, int64_t rows, int64_t cols,
+                          int dt) {
   hipStream_t s = (hipStream_t)stream;
   if (dt == DT_F32)
-    launch_colsum<DT_F32>(s, (const void*)src, (float*)out, rows, cols);
+    launch_colsum<DT_F32>(s, (const void*)src, (float*)out, (float*)scratch, rows, cols);
   else
-    launch_colsum<DT_BF16>(s, (const void*)src, (float*)out, rows, cols);
+    launch_colsum<DT_BF16>(s, (const void*)src, (float*)out, (float*)scratch, rows, cols);
   LAUNCH_CHECK();
   return 0;
 }
 
 SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
                                  uint64_t b, uint64_t dx, uint64_t db,
-                                 int64_t rows, int64_t cols, int dt) {
+                                 uint64_t scratch, int64_t rows, int64_t cols,
+                                 int dt) {
   hipStream_t s = (hipStream_t)stream;
   int64_t n = rows * cols;
   if (cols % 8 == 0) {
@@ -193,9 +216,9 @@ SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
                          (const void*)dy, (const void*)x, (const void*)b, (void*)dx, n, cols);
   }
   if (dt == DT_F32)
-    launch_colsum<DT_F32>(s, (const void*)dx, (float*)db, rows, cols);
+    launch_colsum<DT_F32>(s, (const void*)dx, (float*)db, (float*)scratch, rows, cols);
   else
-    launch_colsum<DT_BF16>(s, (const void*)dx, (float*)db, rows, cols);
+    launch_colsum<DT_BF16>(s, (const void*)dx, (float*)db, (float*)scratch, rows, cols);
   LAUNCH_CHECK();
   return 0;
 }

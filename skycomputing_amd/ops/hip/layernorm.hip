@@ -202,18 +202,21 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_vec_kernel(
   }
 }
 
-// column-parallel dw/db, vectorized: thread owns 8 consecutive columns
-// (16 B loads per row), walks a row slab; 8 atomicAdds per slab.
+// column-parallel dw/db, two-stage, no atomics, outputs need no zero-init:
+// stage 1 writes per-slab partials to scratch [nslabs][2*cols] fp32
+// (dw partial at [y][c], db partial at [y][cols+c]); stage 2 reduces.
+#define LN_SLABS 128
+
 template <int DT, bool HAS_RES, int BLOCK>
-__global__ __launch_bounds__(BLOCK) void ln_bwd_wb_vec_kernel(
+__global__ __launch_bounds__(BLOCK) void ln_bwd_wb_part_kernel(
     const void* __restrict__ dy, const void* __restrict__ x,
     const void* __restrict__ res, const float* __restrict__ mean,
-    const float* __restrict__ rstd, float* __restrict__ dw,
-    float* __restrict__ db, int64_t rows, int64_t cols8,
-    int64_t rows_per_slab) {
+    const float* __restrict__ rstd, float* __restrict__ scratch,
+    int64_t rows, int64_t cols8, int64_t rows_per_slab) {
   const int64_t c8 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
   if (c8 >= cols8) return;
-  const int64_t r0 = (int64_t)blockIdx.y * rows_per_slab;
+  const int64_t slab = blockIdx.y;
+  const int64_t r0 = slab * rows_per_slab;
   const int64_t r1 = min(rows, r0 + rows_per_slab);
   float sw[8] = {0.f}, sb[8] = {0.f};
   for (int64_t r = r0; r < r1; ++r) {
@@ -233,11 +236,27 @@ __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_vec_kernel(
       sb[j] += dyv[j];
     }
   }
+  float* base = scratch + slab * 2 * cols8 * 8;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    atomicAdd(&dw[c8 * 8 + j], sw[j]);
-    atomicAdd(&db[c8 * 8 + j], sb[j]);
+    base[c8 * 8 + j] = sw[j];
+    base[cols8 * 8 + c8 * 8 + j] = sb[j];
   }
+}
+
+template <int BLOCK>
+__global__ __launch_bounds__(BLOCK) void ln_bwd_wb_final_kernel(
+    const float* __restrict__ scratch, float* __restrict__ dw,
+    float* __restrict__ db, int64_t cols, int64_t nslabs) {
+  const int64_t col = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  if (col >= cols) return;
+  float sw = 0.f, sb = 0.f;
+  for (int64_t y = 0; y < nslabs; ++y) {
+    sw += scratch[y * 2 * cols + col];
+    sb += scratch[y * 2 * cols + cols + col];
+  }
+  dw[col] = sw;
+  db[col] = sb;
 }
 
 template <int DT, bool HAS_RES, int BLOCK>
@@ -296,8 +315,8 @@ __global__ __launch_bounds__(BLOCK) void ln_bwd_dx_kernel(
 SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
                                  uint64_t res, uint64_t w, uint64_t mean,
                                  uint64_t rstd, uint64_t dx, uint64_t dw,
-                                 uint64_t db, int64_t rows, int64_t cols,
-                                 int dt) {
+                                 uint64_t db, uint64_t scratch, int64_t rows,
+                                 int64_t cols, int dt) {
   hipStream_t s = (hipStream_t)stream;
   bool has_res = res != 0;
   if (ln_fast_ok(cols)) {
@@ -330,21 +349,24 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
     else              { if (has_res) LNBD(DT_BF16, true); else LNBD(DT_BF16, false); }
 #undef LNBD
   }
-  if (cols % 8 == 0) {
-    constexpr int BLOCK = 256;
+  if (cols % 8 == 0 && scratch != 0) {
+    constexpr int BLOCK = 128;
     const int64_t cols8 = cols / 8;
-    const int64_t cb = (cols8 + BLOCK - 1) / BLOCK;
-    int64_t slab = 16;
-    while (cb * ((rows + slab - 1) / slab) > 1024 && slab < rows) slab *= 2;
-    dim3 grid((unsigned)cb, (unsigned)((rows + slab - 1) / slab));
-#define LNWBV(DT, HR)                                                          \
-  hipLaunchKernelGGL((ln_bwd_wb_vec_kernel<DT, HR, BLOCK>), grid, dim3(BLOCK), \
-                     0, s, (const void*)dy, (const void*)x, (const void*)res,  \
-                     (const float*)mean, (const float*)rstd, (float*)dw,       \
-                     (float*)db, rows, cols8, slab)
-    if (dt == DT_F32) { if (has_res) LNWBV(DT_F32, true); else LNWBV(DT_F32, false); }
-    else              { if (has_res) LNWBV(DT_BF16, true); else LNWBV(DT_BF16, false); }
-#undef LNWBV
+    const int64_t nslabs = rows < LN_SLABS ? rows : LN_SLABS;
+    const int64_t slab = (rows + nslabs - 1) / nslabs;
+    dim3 grid((unsigned)((cols8 + BLOCK - 1) / BLOCK), (unsigned)nslabs);
+#define LNWBP(DT, HR)                                                          \
+  hipLaunchKernelGGL((ln_bwd_wb_part_kernel<DT, HR, BLOCK>), grid,             \
+                     dim3(BLOCK), 0, s, (const void*)dy, (const void*)x,       \
+                     (const void*)res, (const float*)mean,                     \
+                     (const float*)rstd, (float*)scratch, rows, cols8, slab)
+    if (dt == DT_F32) { if (has_res) LNWBP(DT_F32, true); else LNWBP(DT_F32, false); }
+    else              { if (has_res) LNWBP(DT_BF16, true); else LNWBP(DT_BF16, false); }
+#undef LNWBP
+    dim3 g2((unsigned)((cols + 255) / 256));
+    hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256>), g2, dim3(256), 0, s,
+                       (const float*)scratch, (float*)dw, (float*)db, cols,
+                       nslabs);
   } else {
     constexpr int BLOCK = 256;
     // slab sized so the grid fills 256 CUs (cols/256 col-blocks * row-slabs)

@@ -33,12 +33,12 @@ def _register_signatures(lib):
         # layernorm: fwd saves mean/rstd (fp32 per row) for bwd
         "sky_layernorm_fwd": [u64, u64, u64, u64, u64, u64, u64, u64, i64, i64, f32, i32],
         #                     strm  x   res  w    b    y    mean rstd rows cols eps  dt
-        "sky_layernorm_bwd": [u64, u64, u64, u64, u64, u64, u64, u64, u64, u64, i64, i64, i32],
-        #                     strm  dy  x    res  w    mean rstd dx   dw   db   rows cols dt
+        "sky_layernorm_bwd": [u64, u64, u64, u64, u64, u64, u64, u64, u64, u64, u64, i64, i64, i32],
+        #                 strm dy x res w mean rstd dx dw db scratch | rows cols dt
         "sky_bias_gelu_fwd": [u64, u64, u64, u64, i64, i64, i32],
         #                     strm  x    b    y    rows cols dt
-        "sky_bias_gelu_bwd": [u64, u64, u64, u64, u64, u64, i64, i64, i32],
-        #                     strm  dy   x    b    dx   db   rows cols dt
+        "sky_bias_gelu_bwd": [u64, u64, u64, u64, u64, u64, u64, i64, i64, i32],
+        #                 strm dy x b dx db scratch | rows cols dt
         "sky_masked_softmax_fwd": [u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, i32],
         #                     strm scores mask probs B   h    Sq   Sk  scale keep seed dt
         "sky_masked_softmax_bwd": [u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, i32],
@@ -54,7 +54,7 @@ def _register_signatures(lib):
         #                   strm  x    y    n   keep salt state dt
         "sky_dropout_bwd": [u64, u64, u64, i64, f32, u64, u64, i32],
         "sky_rng_tick": [u64, u64],  # strm, state ptr
-        "sky_colsum": [u64, u64, u64, i64, i64, i32],  # strm src out32 rows cols dt
+        "sky_colsum": [u64, u64, u64, u64, i64, i64, i32],  # strm src out32 scratch rows cols dt
         "sky_mfma_probe": [u64, u64, u64, u64],  # strm A B D
         "sky_pack3": [u64, u64, u64, u64, u64, i64, i64, i64, i64, i32],
         "sky_attn_fwd": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, u64],
