@@ -124,15 +124,31 @@ __global__ __launch_bounds__(BLOCK) void colsum_part_kernel(
   for (int j = 0; j < 8; ++j) scratch[(slab * cols8 + c8) * 8 + j] = s[j];
 }
 
+// two-level: block covers 16 columns x 16 slab-groups (each thread sums
+// nslabs/16 slabs of one column), LDS tree, then one write per column.
 template <int BLOCK>
 __global__ __launch_bounds__(BLOCK) void colsum_final_kernel(
     const float* __restrict__ scratch, float* __restrict__ out, int64_t cols,
     int64_t nslabs) {
-  const int64_t col = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
-  if (col >= cols) return;
+  __shared__ float lds[16][17];
+  const int c = threadIdx.x & 15;
+  const int g = threadIdx.x >> 4;
+  const int64_t col = (int64_t)blockIdx.x * 16 + c;
+  const int64_t per = (nslabs + 15) / 16;
   float s = 0.f;
-  for (int64_t y = 0; y < nslabs; ++y) s += scratch[y * cols + col];
-  out[col] = s;
+  if (col < cols) {
+    const int64_t y1 = min(nslabs, (int64_t)(g + 1) * per);
+    for (int64_t y = (int64_t)g * per; y < y1; ++y)
+      s += scratch[y * cols + col];
+  }
+  lds[g][c] = s;
+  __syncthreads();
+  if (g == 0 && col < cols) {
+    float t = 0.f;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) t += lds[j][c];
+    out[col] = t;
+  }
 }
 
 template <int DT, int BLOCK>
@@ -162,7 +178,7 @@ static void launch_colsum(hipStream_t s, const void* src, float* out,
     dim3 grid((unsigned)((cols8 + BLOCK - 1) / BLOCK), (unsigned)nslabs);
     hipLaunchKernelGGL((colsum_part_kernel<DT, BLOCK>), grid, dim3(BLOCK), 0,
                        s, src, scratch, rows, cols8, slab);
-    dim3 g2((unsigned)((cols + 255) / 256));
+    dim3 g2((unsigned)((cols + 15) / 16));
     hipLaunchKernelGGL((colsum_final_kernel<256>), g2, dim3(256), 0, s,
                        scratch, out, cols, nslabs);
     return;

@@ -244,19 +244,36 @@ __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_part_kernel(
   }
 }
 
+// two-level: block covers 16 columns x 16 slab-groups, LDS tree, one
+// write per column (see colsum_final_kernel).
 template <int BLOCK>
 __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_final_kernel(
     const float* __restrict__ scratch, float* __restrict__ dw,
     float* __restrict__ db, int64_t cols, int64_t nslabs) {
-  const int64_t col = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
-  if (col >= cols) return;
+  __shared__ float ldsw[16][17];
+  __shared__ float ldsb[16][17];
+  const int c = threadIdx.x & 15;
+  const int g = threadIdx.x >> 4;
+  const int64_t col = (int64_t)blockIdx.x * 16 + c;
+  const int64_t per = (nslabs + 15) / 16;
   float sw = 0.f, sb = 0.f;
-  for (int64_t y = 0; y < nslabs; ++y) {
-    sw += scratch[y * 2 * cols + col];
-    sb += scratch[y * 2 * cols + cols + col];
+  if (col < cols) {
+    const int64_t y1 = min(nslabs, (int64_t)(g + 1) * per);
+    for (int64_t y = (int64_t)g * per; y < y1; ++y) {
+      sw += scratch[y * 2 * cols + col];
+      sb += scratch[y * 2 * cols + cols + col];
+    }
   }
-  dw[col] = sw;
-  db[col] = sb;
+  ldsw[g][c] = sw;
+  ldsb[g][c] = sb;
+  __syncthreads();
+  if (g == 0 && col < cols) {
+    float tw = 0.f, tb = 0.f;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) { tw += ldsw[j][c]; tb += ldsb[j][c]; }
+    dw[col] = tw;
+    db[col] = tb;
+  }
 }
 
 template <int DT, bool HAS_RES, int BLOCK>
@@ -363,7 +380,7 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
     if (dt == DT_F32) { if (has_res) LNWBP(DT_F32, true); else LNWBP(DT_F32, false); }
     else              { if (has_res) LNWBP(DT_BF16, true); else LNWBP(DT_BF16, false); }
 #undef LNWBP
-    dim3 g2((unsigned)((cols + 255) / 256));
+    dim3 g2((unsigned)((cols + 15) / 16));
     hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256>), g2, dim3(256), 0, s,
                        (const float*)scratch, (float*)dw, (float*)db, cols,
                        nslabs);
