@@ -1,0 +1,77 @@
+"""Canonical experiment config: BERT on MNLI-shaped data, SPMD pipeline.
+
+Mirrors the reference experiment config's shape and keys
+(reference: experiment/config.py:1-167) adapted to the SPMD world: no
+rpc_config (RCCL init comes from torchrun env), worker extra_configs index
+by rank, layer list built by the same registry IR.
+
+LAYER_NUM follows the reference convention: the number of encoder layers;
+each becomes a Head/Body/Tail triplet in the pipeline layer list.
+"""
+
+import os
+
+# ---- model geometry (BERT-large: H=1024, A=16; reference config.py:11-17) ----
+LAYER_NUM = int(os.environ.get("SKY_LAYER_NUM", 24))
+NUM_CLASS = 3
+
+BERT_CONFIG = dict(
+    vocab_size=30522,
+    hidden_size=1024,
+    num_attention_heads=16,
+    intermediate_size=4096,
+    max_position_embeddings=512,
+    hidden_dropout_prob=0.1,
+    attention_probs_dropout_prob=0.1,
+)
+
+model_config = dict(
+    kind="bert",
+    num_encoder_layers=LAYER_NUM,
+    bert_config=BERT_CONFIG,
+    num_class=NUM_CLASS,
+)
+
+# ---- data (reference config.py:105-120) ----
+data_config = dict(
+    batch_size=32,
+    dataset=dict(
+        layer_type="SyntheticGlueDataset",  # offline image: synthetic MNLI shapes
+        size=4096,
+        max_seq_length=128,
+        vocab_size=BERT_CONFIG["vocab_size"],
+        num_class=NUM_CLASS,
+        seed=7,
+    ),
+)
+
+# ---- per-rank worker knobs (reference config.py:75-99) ----
+# slowdown > 0 simulates a slow device (GPU busy-spin scaled by measured
+# stage time); mem_limit caps the allocator's view of free memory.
+worker_config = [
+    dict(slowdown=0.0, mem_limit=None),
+]  # extended/cycled to world size by launch.py
+
+# ---- allocation (reference config.py:123-151) ----
+allocator_config = dict(
+    mode=os.environ.get("ALLOCATE_TYPE", "even"),  # even | dynamic | optimal
+    benchmark=dict(batch_size=32, seq_len=128, iterations=5),
+    stimulate=os.environ.get("STIMULATE") == "1",
+)
+
+# ---- training (reference config.py:154-167) ----
+train_config = dict(
+    max_epoch=1,
+    max_iter=30,
+    optimizer=dict(lr=1e-3, momentum=0.0, weight_decay=0.0),
+    num_microbatches=int(os.environ.get("SKY_MICROBATCHES", 0)),  # 0 = auto
+    schedule="gpipe",
+    log_interval=1,
+    hooks=[
+        dict(layer_type="TimerHook"),
+        dict(layer_type="StopHook", root="."),
+        # dict(layer_type="CheckpointHook", save_path="./checkpoints", save_interval=1),
+    ],
+)
+
+logging_config = dict(log_dir="./logs")

@@ -30,7 +30,9 @@ from .estimator import Estimator
 from .worker import WorkerManager
 
 
-def default_bert_probe_cfg(hidden: int = 1024, heads: int = 16, intermediate: int = 4096):
+def default_bert_probe_cfg(hidden: int = 1024, heads: int | None = None, intermediate: int | None = None):
+    heads = heads or max(1, hidden // 64)
+    intermediate = intermediate or 4 * hidden
     bc = dict(hidden_size=hidden, num_attention_heads=heads, intermediate_size=intermediate,
               hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0)
     return [
