@@ -38,7 +38,7 @@ DEV void glds16(const void* g, void* l) {
 // epilogue modes
 enum { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_GELU = 2 };
 
-template <bool A_DIRECT, bool B_DIRECT, int EPI, bool STORE_Z>
+template <bool A_DIRECT, bool B_DIRECT, int EPI, bool STORE_Z, bool USE_GLDS>
 __global__ __launch_bounds__(GM_BLOCK) void gemm_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
     ushort_t* __restrict__ C, const ushort_t* __restrict__ bias,
@@ -76,7 +76,10 @@ __global__ __launch_bounds__(GM_BLOCK) void gemm_kernel(
       const int u = it * 256 + tid;
       const int row = u >> 3, c16 = u & 7;
       const ushort_t* g = src + (size_t)(row0 + (row < rows ? row : rows - 1)) * ld + k0 + c16 * 8;
-      glds16(g, dst + u * 16);
+      if (USE_GLDS)
+        glds16(g, dst + u * 16);
+      else
+        *(ushort8_t*)(dst + u * 16) = *(const ushort8_t*)g;
     }
   };
   auto stage_trans = [&](const ushort_t* src, int ld, int k0, int col0,
@@ -115,7 +118,7 @@ __global__ __launch_bounds__(GM_BLOCK) void gemm_kernel(
 
   const int NT = K / GM_BK;
   stage(0, 0);
-  if ((A_DIRECT || B_DIRECT))
+  if (USE_GLDS && (A_DIRECT || B_DIRECT))
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -142,7 +145,8 @@ __global__ __launch_bounds__(GM_BLOCK) void gemm_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j], acc[i][j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (USE_GLDS && (A_DIRECT || B_DIRECT))
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
   }
 
@@ -179,7 +183,8 @@ __global__ __launch_bounds__(GM_BLOCK) void gemm_kernel(
 SKY_EXPORT int sky_gemm(uint64_t stream, uint64_t A, uint64_t B, uint64_t C,
                         uint64_t bias, uint64_t Z, int64_t M, int64_t N,
                         int64_t K, int64_t lda, int64_t ldb, int64_t ldc,
-                        int transA, int transB, int epi, int dt) {
+                        int transA, int transB, int epi, int dt,
+                        int use_glds) {
   if (dt != DT_BF16) return (int)hipErrorInvalidValue;
   if (M % GM_BM || N % GM_BN || K % GM_BK) return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
@@ -188,10 +193,20 @@ SKY_EXPORT int sky_gemm(uint64_t stream, uint64_t A, uint64_t B, uint64_t C,
   const bool ad = transA == 0, bd = transB != 0;
   const bool sz = Z != 0;
 #define GK(AD, BD, EP, SZ)                                                     \
-  hipLaunchKernelGGL((gemm_kernel<AD, BD, EP, SZ>), grid, dim3(GM_BLOCK),      \
-                     lds_bytes, s, (const ushort_t*)A, (const ushort_t*)B,     \
-                     (ushort_t*)C, (const ushort_t*)bias, (ushort_t*)Z,        \
-                     (int)M, (int)N, (int)K, (int)lda, (int)ldb, (int)ldc)
+  do {                                                                         \
+    if (use_glds)                                                              \
+      hipLaunchKernelGGL((gemm_kernel<AD, BD, EP, SZ, true>), grid,            \
+                         dim3(GM_BLOCK), lds_bytes, s, (const ushort_t*)A,     \
+                         (const ushort_t*)B, (ushort_t*)C,                     \
+                         (const ushort_t*)bias, (ushort_t*)Z, (int)M, (int)N,  \
+                         (int)K, (int)lda, (int)ldb, (int)ldc);                \
+    else                                                                       \
+      hipLaunchKernelGGL((gemm_kernel<AD, BD, EP, SZ, false>), grid,           \
+                         dim3(GM_BLOCK), lds_bytes, s, (const ushort_t*)A,     \
+                         (const ushort_t*)B, (ushort_t*)C,                     \
+                         (const ushort_t*)bias, (ushort_t*)Z, (int)M, (int)N,  \
+                         (int)K, (int)lda, (int)ldb, (int)ldc);                \
+  } while (0)
 #define GK_EPI(AD, BD)                                                         \
   do {                                                                         \
     if (epi == EPI_NONE) GK(AD, BD, EPI_NONE, false);                          \

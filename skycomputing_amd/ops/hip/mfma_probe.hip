@@ -41,3 +41,35 @@ SKY_EXPORT int sky_mfma_probe(uint64_t stream, uint64_t A, uint64_t B,
   LAUNCH_CHECK();
   return 0;
 }
+
+// glds semantics probe: copy 4 KB global -> LDS via global_load_lds with a
+// per-lane (mode 0) or wave-uniform (mode 1) LDS pointer, then write the
+// LDS back out so the host can check which addressing lands correctly.
+typedef const __attribute__((address_space(1))) unsigned int* gp_u32;
+typedef __attribute__((address_space(3))) unsigned int* lp_u32;
+
+__global__ __launch_bounds__(256) void glds_probe_kernel(
+    const ushort_t* __restrict__ src, ushort_t* __restrict__ dst, int mode) {
+  __shared__ __attribute__((aligned(16))) char lbuf[4096];
+  const int tid = threadIdx.x;
+  const int l = tid & 63;
+  const int w = tid >> 6;
+  const ushort_t* g = src + tid * 8;
+  if (mode == 0) {
+    __builtin_amdgcn_global_load_lds((gp_u32)g, (lp_u32)(lbuf + tid * 16), 16, 0, 0);
+  } else {
+    __builtin_amdgcn_global_load_lds((gp_u32)g, (lp_u32)(lbuf + w * 64 * 16), 16, 0, 0);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  *(ushort8_t*)(dst + tid * 8) = *(const ushort8_t*)(lbuf + tid * 16);
+}
+
+SKY_EXPORT int sky_glds_probe(uint64_t stream, uint64_t src, uint64_t dst,
+                              int mode) {
+  hipLaunchKernelGGL(glds_probe_kernel, dim3(1), dim3(256), 0,
+                     (hipStream_t)stream, (const ushort_t*)src,
+                     (ushort_t*)dst, mode);
+  LAUNCH_CHECK();
+  return 0;
+}

@@ -1,0 +1,59 @@
+#!/usr/bin/env python3
+"""Standalone timing of the fused attention kernels at the bench shape."""
+
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, "/root/repo")
+from skycomputing_amd.ops import hiplib
+from skycomputing_amd.ops.hiplib import check, ptr
+
+lib = hiplib.require()
+torch.manual_seed(0)
+B, S, h, d = 32, 128, 16, 64
+qkv = torch.randn(B, S, 3, h, d, dtype=torch.bfloat16, device="cuda")
+mask = torch.zeros(B, 1, 1, S, dtype=torch.bfloat16, device="cuda")
+out = torch.empty(B, S, h, d, dtype=torch.bfloat16, device="cuda")
+m = torch.empty(B, h, S, dtype=torch.float32, device="cuda")
+l = torch.empty_like(m)
+P = torch.empty(B, h, S, S, dtype=torch.bfloat16, device="cuda")
+stream = torch.cuda.current_stream().cuda_stream
+scale = 0.125
+
+
+def run_fwd(keep=0.9):
+    check(lib.sky_attn_fwd(stream, ptr(qkv), ptr(mask), ptr(out), ptr(m),
+                           ptr(l), B, S, h, d, scale, keep, 123, 0), "fwd")
+
+
+def run_probs(keep=0.9):
+    check(lib.sky_attn_probs(stream, ptr(qkv), ptr(mask), ptr(m), ptr(l),
+                             ptr(P), ptr(P), B, S, h, d, scale, keep, 123, 0),
+          "probs")
+
+
+def bench(fn, iters=50):
+    for _ in range(5):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e6
+
+
+run_fwd()  # populate m/l for probs mode
+torch.cuda.synchronize()
+print(f"attn_fwd  keep=0.9: {bench(lambda: run_fwd(0.9)):8.1f} us")
+print(f"attn_fwd  keep=1.0: {bench(lambda: run_fwd(1.0)):8.1f} us")
+print(f"attn_probs keep=.9: {bench(lambda: run_probs(0.9)):8.1f} us")
+
+# reference: torch sdpa on the same problem
+q = qkv[:, :, 0].permute(0, 2, 1, 3).contiguous()
+k = qkv[:, :, 1].permute(0, 2, 1, 3).contiguous()
+v = qkv[:, :, 2].permute(0, 2, 1, 3).contiguous()
+print(f"torch sdpa (no drop): "
+      f"{bench(lambda: torch.nn.functional.scaled_dot_product_attention(q, k, v)):8.1f} us")

@@ -12,6 +12,10 @@ from skycomputing_amd.ops import hiplib
 from skycomputing_amd.ops.hiplib import check, ptr
 
 
+import os
+USE_GLDS = int(os.environ.get('SKY_GEMM_GLDS', '1'))
+
+
 def sky_gemm(a, b, bias=None, transA=0, transB=1, epi=0, z=None):
     lib = hiplib.require()
     if transA == 0:
@@ -27,7 +31,7 @@ def sky_gemm(a, b, bias=None, transA=0, transB=1, epi=0, z=None):
         lib.sky_gemm(
             torch.cuda.current_stream().cuda_stream, ptr(a), ptr(b), ptr(c),
             ptr(bias), ptr(z), M, N, K, a.stride(0), b.stride(0), c.stride(0),
-            transA, transB, epi, 1,
+            transA, transB, epi, 1, USE_GLDS,
         ),
         "sky_gemm",
     )
