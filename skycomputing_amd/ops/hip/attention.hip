@@ -24,6 +24,9 @@
 
 #include "common.h"
 
+typedef const __attribute__((address_space(1))) unsigned int* att_gas;
+typedef __attribute__((address_space(3))) unsigned int* att_las;
+
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
@@ -61,6 +64,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const uint64_t seed =
       salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull;
   const float inv_keep = 1.f / keep;
+  const unsigned keep16 = keep_to_16(keep);
 
   const ushort_t* qbase = qkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
   const ushort_t* kbase = qbase + (size_t)h * ATT_D;
@@ -75,12 +79,15 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     // P tiles still need zeroing for partial last q tile? S==128 -> full.
   }
 
-  // ---- stage K into LDS [S][64] (swizzled rows of 128 B) ----
+  // ---- stage K into LDS [S][64] via global_load_lds: the XOR swizzle
+  // moves to the SOURCE address (lane-linear LDS image == swizzled layout;
+  // guide rule 21) ----
   for (int u = tid; u < S * 8; u += 256) {
     const int tok = u >> 3;
-    const int c16 = u & 7;
-    ushort8_t v = *(const ushort8_t*)(kbase + (size_t)tok * ts + c16 * 8);
-    *(ushort8_t*)lds_at(lds, swz(K_OFF + tok * 128 + c16 * 16, tok, 7)) = v;
+    const int c16s = (u & 7) ^ (tok & 7);  // pre-swizzled source column
+    __builtin_amdgcn_global_load_lds(
+        (att_gas)(kbase + (size_t)tok * ts + c16s * 8),
+        (att_las)lds_at(lds, K_OFF + u * 16), 16, 0, 0);
   }
   // ---- stage V transposed into LDS [64][S] ----
   for (int u = tid; u < S * 8; u += 256) {
@@ -93,9 +100,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       *(ushort_t*)lds_at(lds, swz(VT_OFF + c * 256 + tok * 2, c, 15)) = v[j];
     }
   }
-  __syncthreads();
-
-  // ---- per-wave: 32 q rows = q tiles {2w, 2w+1} ----
+  // ---- per-wave constants (issued BEFORE the barrier so the global
+  // loads land under the staging latency) ----
   const int qt0 = 2 * w;
   const int lm = l & 15;
   const int lg = l >> 4;  // lane group 0..3
@@ -109,6 +115,17 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     if (HAS_MASK && col < S) mv = bf16_to_f32(mask[(size_t)b * S + col]);
     mval[kt] = (col < S) ? mv : -3.0e38f;
   }
+  // Q fragments for both q tiles (independent of LDS)
+  bf16x8 aq_all[2][2];
+#pragma unroll
+  for (int qi = 0; qi < 2; ++qi) {
+    int qtok = (qt0 + qi) * 16 + lm;
+    if (qtok >= S) qtok = S - 1;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)
+      aq_all[qi][ks] = *(const bf16x8*)(qbase + (size_t)qtok * ts + ks * 32 + lg * 8);
+  }
+  __syncthreads();
 
   f32x4 oacc[2][4];
 #pragma unroll
@@ -120,15 +137,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const int qtile = qt0 + qi;
     const int qtok_base = qtile * 16;
     if (qtok_base >= S) break;
-    // A fragments of Q (two K=32 splits of d=64)
-    bf16x8 aq[2];
-    {
-      int qtok = qtok_base + lm;
-      if (qtok >= S) qtok = S - 1;  // clamped read; row skipped at write
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
-        aq[ks] = *(const bf16x8*)(qbase + (size_t)qtok * ts + ks * 32 + lg * 8);
-    }
+    // A fragments of Q (preloaded before the barrier)
+    bf16x8 aq[2] = {aq_all[qi][0], aq_all[qi][1]};
     f32x4 sacc[8];
 #pragma unroll
     for (int kt = 0; kt < 8; ++kt) sacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
@@ -206,19 +216,29 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         }
       }
     }
-    // normalize, dropout, write P (LDS in fwd mode; global in probs mode)
+    // normalize, dropout, write P (LDS in fwd mode; global in probs mode).
+    // Dropout indices are lane-local (base*8 + kt), so one hash covers 4
+    // consecutive kt tiles; backward reconstructs the mask from Pd != 0
+    // (FusedAttentionFn.backward), so the index scheme is kernel-private.
 #pragma unroll
-    for (int kt = 0; kt < 8; ++kt) {
-      if (kt >= NT) continue;
-      const int col = kt * 16 + lm;
+    for (int r = 0; r < 4; ++r) {
+      const int row = qtok_base + lg * 4 + r;
+      uint64_t z0 = 0, z1 = 0;
+      if (keep < 1.f) {
+        const uint64_t base = ((uint64_t)bh * S + row) * 16 + lm;
+        z0 = rng_hash(seed, base * 2);
+        z1 = rng_hash(seed, base * 2 + 1);
+      }
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = qtok_base + lg * 4 + r;
+      for (int kt = 0; kt < 8; ++kt) {
+        if (kt >= NT) continue;
+        const int col = kt * 16 + lm;
         float p = z[kt][r] * inv[r];
         float pd = p;
         if (keep < 1.f) {
-          const uint64_t idx = ((uint64_t)bh * S + row) * S + col;
-          pd = (rng_uniform(seed, idx) < keep) ? p * inv_keep : 0.f;
+          const uint64_t zz = kt < 4 ? z0 : z1;
+          bool kbit = (unsigned)((zz >> (16 * (kt & 3))) & 0xFFFFu) < keep16;
+          pd = kbit ? p * inv_keep : 0.f;
         }
         if (PROBS_MODE) {
           if (row < S && col < S) {

@@ -121,12 +121,27 @@ DEV float block_sum(float v, float* lds /* >= BLOCK/64 floats */) {
 }
 
 // counter-based RNG: splitmix64 -> uniform [0,1)
-DEV float rng_uniform(uint64_t seed, uint64_t idx) {
-  uint64_t z = seed + idx * 0x9E3779B97F4A7C15ull;
+DEV uint64_t rng_hash(uint64_t seed, uint64_t q) {
+  uint64_t z = seed + q * 0x9E3779B97F4A7C15ull;
   z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
   z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
-  z = z ^ (z >> 31);
-  return (float)(z >> 40) * 0x1.0p-24f;
+  return z ^ (z >> 31);
+}
+
+DEV float rng_uniform(uint64_t seed, uint64_t idx) {
+  return (float)(rng_hash(seed, idx) >> 40) * 0x1.0p-24f;
+}
+
+// dropout keep decision: one splitmix hash feeds FOUR consecutive indices
+// (16-bit thresholds) — 4x fewer hashes when a lane owns consecutive
+// elements. keep16 = keep * 65536 (keep == 1 -> always true).
+DEV bool rng_keep16(uint64_t seed, uint64_t idx, unsigned keep16) {
+  uint64_t z = rng_hash(seed, idx >> 2);
+  return (unsigned)((z >> (16 * (idx & 3))) & 0xFFFFu) < keep16;
+}
+
+DEV unsigned keep_to_16(float keep) {
+  return keep >= 1.f ? 0x20000u : (unsigned)(keep * 65536.f + 0.5f);
 }
 
 // erf-formula GELU and its derivative (fp32)

@@ -254,14 +254,19 @@ __global__ __launch_bounds__(256) void dropout_vec_kernel(
     const void* __restrict__ x, void* __restrict__ y, int64_t n8, float keep,
     uint64_t salt, const unsigned long long* __restrict__ state) {
   const float inv_keep = 1.f / keep;
+  const unsigned keep16 = keep_to_16(keep);
   const uint64_t seed = salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull;
   for (int64_t i8 = (int64_t)blockIdx.x * 256 + threadIdx.x; i8 < n8;
        i8 += (int64_t)gridDim.x * 256) {
     float v[8];
     Vec8<DT>::load(x, i8, v);
+    // 8 consecutive elements = exactly 2 hash quads
+    const uint64_t z0 = rng_hash(seed, (uint64_t)i8 * 2);
+    const uint64_t z1 = rng_hash(seed, (uint64_t)i8 * 2 + 1);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      bool k = rng_uniform(seed, (uint64_t)(i8 * 8 + j)) < keep;
+      const uint64_t z = j < 4 ? z0 : z1;
+      bool k = (unsigned)((z >> (16 * (j & 3))) & 0xFFFFu) < keep16;
       v[j] = k ? v[j] * inv_keep : 0.f;
     }
     Vec8<DT>::store(y, i8, v);
@@ -277,7 +282,7 @@ __global__ __launch_bounds__(256) void dropout_kernel(
   for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * 256) {
     float v = load_elem<DT>(x, i);
-    bool k = rng_uniform(seed, (uint64_t)i) < keep;
+    bool k = rng_keep16(seed, (uint64_t)i, keep_to_16(keep));
     store_elem<DT>(y, i, k ? v * inv_keep : 0.f);
   }
 }

@@ -26,7 +26,9 @@ def tiny_bert_cfg(num_encoder_layers: int = 2, dropout: float = 0.0) -> list[dic
 
 def run_multiprocess(fn, world_size: int, port: int, *args, timeout: float = 180.0):
     """Spawn `world_size` ranks running fn(rank, world_size, *args); raise on
-    any failure."""
+    any failure. The rendezvous port is jittered by pid to avoid TIME_WAIT
+    collisions between test runs."""
+    port = 20000 + (port + os.getpid() * 7) % 40000
     ctx = mp.get_context("spawn")
     procs = []
     for rank in range(world_size):
