@@ -113,5 +113,31 @@ class Runner:
                 break
         self.call_hook("after_run")
 
+    def val(self, data_loader, max_batches: int | None = None) -> float | None:
+        """Forward-only evaluation; returns classification accuracy
+        (broadcast to all ranks). The reference exposed val hooks but no
+        loop (scaelum/runner/hooks.py:5-58); this completes the lifecycle."""
+        self.engine.eval()
+        self.call_hook("before_val_epoch")
+        correct = total = 0
+        for bi, (data, labels) in enumerate(data_loader):
+            if max_batches is not None and bi >= max_batches:
+                break
+            self.call_hook("before_val_iter")
+            logits = self.engine.evaluate_batch(data, labels)
+            if logits is not None:
+                pred = logits.float().argmax(-1).cpu()
+                correct += int((pred == labels).sum())
+                total += labels.numel()
+            self.call_hook("after_val_iter")
+        acc = (correct / total) if total else None
+        src = self.engine.plan.stage_ranks[-1] if self.engine.plan.num_stages else 0
+        acc = self.comm.broadcast_object(acc, src=src)
+        self.call_hook("after_val_epoch")
+        self.engine.train(True)
+        if acc is not None:
+            self.logger.info(f"val accuracy {acc:.4f}")
+        return acc
+
     def stop(self):
         self.should_stop = True
