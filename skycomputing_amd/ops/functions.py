@@ -8,6 +8,8 @@ fp32 (tests/test_ops_gpu.py).
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 from . import hiplib
@@ -277,6 +279,23 @@ class FusedAttentionFn(torch.autograd.Function):
         qkv, mask, m, lsum = ctx.saved_tensors
         B, S, _, h, d = qkv.shape
         dev = qkv.device
+        if os.environ.get("SKY_NO_FUSED_ATTN_BWD") != "1":
+            # fully-fused path: two MFMA kernels, no torch ops at all
+            dout = dout.contiguous()
+            alloc = torch.empty if S == 128 else torch.zeros
+            pdT = alloc((B, h, S, S), dtype=qkv.dtype, device=dev)
+            dsT = alloc((B, h, S, S), dtype=qkv.dtype, device=dev)
+            dqkv = torch.empty_like(qkv)
+            check(
+                lib.sky_attn_bwd(
+                    _stream(), ptr(qkv), ptr(dout), ptr(mask), ptr(m),
+                    ptr(lsum), ptr(pdT), ptr(dsT), ptr(dqkv),
+                    B, S, h, d, ctx.scale, ctx.keep, ctx.salt,
+                    rng_state().data_ptr(),
+                ),
+                "sky_attn_bwd",
+            )
+            return dqkv, None, None, None, None
         P = torch.empty(B, h, S, S, dtype=qkv.dtype, device=dev)
         Pd = torch.empty_like(P) if ctx.keep < 1.0 else P
         check(

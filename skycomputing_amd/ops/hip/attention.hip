@@ -383,3 +383,326 @@ SKY_EXPORT int sky_attn_probs(uint64_t stream, uint64_t qkv, uint64_t mask,
   return attn_launch(stream, qkv, mask, 0, m, lsum, p, pd, B, S, h, d, scale,
                      keep, salt, state, true);
 }
+
+// ============================================================================
+// Fully-fused attention backward (two kernels).
+//
+// B1, per (b,h): recompute scores -> P (saved m/l) + dropout mask; compute
+//   dPd = dO V^T (MFMA, A = dO straight from global), dP, the softmax
+//   row-dot, dS = scale*P*(dP - rowdot); write dS row-major to per-wave LDS
+//   and dS^T / Pd^T TRANSPOSED to global scratch; compute
+//   dQ = dS K (A = dS from LDS, B = K^T from LDS) -> dqkv[:,:,0].
+// B2, per (b,h): dV = Pd^T dO and dK = dS^T Q, reading the transposed
+//   scratch directly as MFMA A-fragments (contiguous q per lane) and
+//   staging Q^T / dO^T in LDS -> dqkv[:,:,1:3].
+//
+// Replaces: sky_attn_probs + 4 hipBLASLt bmms + softmax-bwd + dropout-bwd
+// + permute copies + the dqkv pack (FusedAttentionFn.backward fast path).
+// ============================================================================
+
+template <bool HAS_MASK>
+__global__ __launch_bounds__(256) void attn_bwd1_kernel(
+    const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ dout,
+    const ushort_t* __restrict__ mask, const float* __restrict__ m_io,
+    const float* __restrict__ l_io, ushort_t* __restrict__ pdT,
+    ushort_t* __restrict__ dsT, ushort_t* __restrict__ dqkv, int B, int S,
+    int h, float scale, float keep, uint64_t salt,
+    const unsigned long long* __restrict__ state) {
+  // LDS: V [128][64] sw7 @0 (16K), K [128][64] sw7 @16K, Kt [64][128] sw15
+  // @32K, per-wave dSrow [32][128] sw15 @48K+w*8K. Total 80K.
+  extern __shared__ __attribute__((aligned(16))) char lds[];
+  const int tid = threadIdx.x;
+  const int l = tid & 63;
+  const int w = tid >> 6;
+  const int lm = l & 15;
+  const int lg = l >> 4;
+  const int bh = blockIdx.x;
+  const int b = bh / h;
+  const int hh = bh % h;
+  const int ts = 3 * h * ATT_D;
+  const int NT = (S + 15) / 16;
+  const uint64_t seed = salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull;
+  const float inv_keep = 1.f / keep;
+  const unsigned keep16 = keep_to_16(keep);
+
+  const ushort_t* qbase = qkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
+  const ushort_t* kbase = qbase + (size_t)h * ATT_D;
+  const ushort_t* vbase = qbase + (size_t)2 * h * ATT_D;
+  const ushort_t* dobase = dout + ((size_t)b * S * h + hh) * ATT_D;
+  const int dots = h * ATT_D;  // dout token stride
+
+  if (S < ATT_SMAX) {
+    for (int u = tid; u < (80 * 1024) / 16; u += 256)
+      *(ushort8_t*)lds_at(lds, u * 16) = (ushort8_t)(ushort_t)0;
+    __syncthreads();
+  }
+  // stage V and K [tok][64] sw7, Kt [c][tok] sw15
+  for (int u = tid; u < S * 8; u += 256) {
+    const int tok = u >> 3;
+    const int c16 = u & 7;
+    ushort8_t v = *(const ushort8_t*)(vbase + (size_t)tok * ts + c16 * 8);
+    *(ushort8_t*)lds_at(lds, swz(0 + tok * 128 + c16 * 16, tok, 7)) = v;
+    ushort8_t kv = *(const ushort8_t*)(kbase + (size_t)tok * ts + c16 * 8);
+    *(ushort8_t*)lds_at(lds, swz(16384 + tok * 128 + c16 * 16, tok, 7)) = kv;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c16 * 8 + j;
+      *(ushort_t*)lds_at(lds, swz(32768 + c * 256 + tok * 2, c, 15)) = kv[j];
+    }
+  }
+  const int qt0 = 2 * w;
+  // preload Q and dO fragments + mask before the barrier
+  float mval[8];
+#pragma unroll
+  for (int kt = 0; kt < 8; ++kt) {
+    const int col = kt * 16 + lm;
+    float mv = 0.f;
+    if (HAS_MASK && col < S) mv = bf16_to_f32(mask[(size_t)b * S + col]);
+    mval[kt] = (col < S) ? mv : -3.0e38f;
+  }
+  bf16x8 aq_all[2][2], ado_all[2][2];
+#pragma unroll
+  for (int qi = 0; qi < 2; ++qi) {
+    int qtok = (qt0 + qi) * 16 + lm;
+    if (qtok >= S) qtok = S - 1;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      aq_all[qi][ks] = *(const bf16x8*)(qbase + (size_t)qtok * ts + ks * 32 + lg * 8);
+      ado_all[qi][ks] = *(const bf16x8*)(dobase + (size_t)qtok * dots + ks * 32 + lg * 8);
+    }
+  }
+  __syncthreads();
+
+  for (int qi = 0; qi < 2; ++qi) {
+    const int qtok_base = (qt0 + qi) * 16;
+    if (qtok_base >= S) break;
+    f32x4 sacc[8], dacc[8];
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      sacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
+      dacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      if (kt >= NT) continue;
+      const int ktok = kt * 16 + lm;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 bk = *(const bf16x8*)lds_at(
+            lds, swz(16384 + ktok * 128 + (ks * 32 + lg * 8) * 2, ktok, 7));
+        sacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq_all[qi][ks], bk, sacc[kt], 0, 0, 0);
+        bf16x8 bv = *(const bf16x8*)lds_at(
+            lds, swz(0 + ktok * 128 + (ks * 32 + lg * 8) * 2, ktok, 7));
+        dacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado_all[qi][ks], bv, dacc[kt], 0, 0, 0);
+      }
+    }
+    // P, Pd, dP, rowdot, dS  (rows r of this lane: qtok_base + lg*4 + r)
+    float mrow[4], lrow[4], dot[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qtok_base + lg * 4 + r;
+      const int rr = row < S ? row : S - 1;
+      mrow[r] = m_io[(size_t)bh * S + rr];
+      lrow[r] = 1.f / l_io[(size_t)bh * S + rr];
+      dot[r] = 0.f;
+    }
+    float pP[8][4], pPd[8][4], pdP[8][4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qtok_base + lg * 4 + r;
+      uint64_t z0 = 0, z1 = 0;
+      if (keep < 1.f) {
+        const uint64_t base = ((uint64_t)bh * S + row) * 16 + lm;
+        z0 = rng_hash(seed, base * 2);
+        z1 = rng_hash(seed, base * 2 + 1);
+      }
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        if (kt >= NT) continue;
+        float p = __expf(sacc[kt][r] * scale + mval[kt] - mrow[r]) * lrow[r];
+        const int col = kt * 16 + lm;
+        if (!(row < S && col < S)) p = 0.f;
+        float pd = p, dp = dacc[kt][r];
+        if (keep < 1.f) {
+          const uint64_t zz = kt < 4 ? z0 : z1;
+          bool kbit = (unsigned)((zz >> (16 * (kt & 3))) & 0xFFFFu) < keep16;
+          pd = kbit ? p * inv_keep : 0.f;
+          dp = kbit ? dp * inv_keep : 0.f;
+        }
+        pP[kt][r] = p;
+        pPd[kt][r] = pd;
+        pdP[kt][r] = dp;
+        dot[r] += dp * p;
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) dot[r] += __shfl_xor(dot[r], off, 64);
+    }
+    // dS = scale * P * (dP - dot); write dSrow (LDS) + dS^T, Pd^T (global)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qtok_base + lg * 4 + r;
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        if (kt >= NT) continue;
+        const int col = kt * 16 + lm;
+        const float ds = scale * pP[kt][r] * (pdP[kt][r] - dot[r]);
+        const int rl = qi * 16 + lg * 4 + r;
+        *(ushort_t*)lds_at(lds, swz(49152 + w * 8192 + rl * 256 + col * 2, rl, 15)) =
+            f32_to_bf16(ds);
+        if (row < S && col < S) {
+          dsT[((size_t)bh * S + col) * S + row] = f32_to_bf16(ds);
+          pdT[((size_t)bh * S + col) * S + row] = f32_to_bf16(pPd[kt][r]);
+        }
+      }
+    }
+    // dQ = dS K : A = dSrow (this wave's LDS), B = Kt
+    f32x4 qacc[4];
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) qacc[ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      if (ks * 32 >= S) break;
+      const int rl = qi * 16 + lm;
+      bf16x8 asr = *(const bf16x8*)lds_at(
+          lds, swz(49152 + w * 8192 + rl * 256 + (ks * 32 + lg * 8) * 2, rl, 15));
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        const int c = ct * 16 + lm;
+        bf16x8 bkt = *(const bf16x8*)lds_at(
+            lds, swz(32768 + c * 256 + (ks * 32 + lg * 8) * 2, c, 15));
+        qacc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(asr, bkt, qacc[ct], 0, 0, 0);
+      }
+    }
+    ushort_t* dq = dqkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qtok_base + lg * 4 + r;
+        const int c = ct * 16 + lm;
+        if (row < S) dq[(size_t)row * ts + c] = f32_to_bf16(qacc[ct][r]);
+      }
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void attn_bwd2_kernel(
+    const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ dout,
+    const ushort_t* __restrict__ pdT, const ushort_t* __restrict__ dsT,
+    ushort_t* __restrict__ dqkv, int B, int S, int h) {
+  // LDS: Qt [64][128] sw15 @0 (16K), dOt [64][128] sw15 @16K. Total 32K.
+  extern __shared__ __attribute__((aligned(16))) char lds[];
+  const int tid = threadIdx.x;
+  const int l = tid & 63;
+  const int w = tid >> 6;
+  const int lm = l & 15;
+  const int lg = l >> 4;
+  const int bh = blockIdx.x;
+  const int b = bh / h;
+  const int hh = bh % h;
+  const int ts = 3 * h * ATT_D;
+  const int dots = h * ATT_D;
+
+  const ushort_t* qbase = qkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
+  const ushort_t* dobase = dout + ((size_t)b * S * h + hh) * ATT_D;
+
+  if (S < ATT_SMAX) {
+    for (int u = tid; u < (32 * 1024) / 16; u += 256)
+      *(ushort8_t*)lds_at(lds, u * 16) = (ushort8_t)(ushort_t)0;
+    __syncthreads();
+  }
+  for (int u = tid; u < S * 8; u += 256) {
+    const int tok = u >> 3;
+    const int c16 = u & 7;
+    ushort8_t qv = *(const ushort8_t*)(qbase + (size_t)tok * ts + c16 * 8);
+    ushort8_t dv = *(const ushort8_t*)(dobase + (size_t)tok * dots + c16 * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c16 * 8 + j;
+      *(ushort_t*)lds_at(lds, swz(0 + c * 256 + tok * 2, c, 15)) = qv[j];
+      *(ushort_t*)lds_at(lds, swz(16384 + c * 256 + tok * 2, c, 15)) = dv[j];
+    }
+  }
+  __syncthreads();
+
+  const int kt0 = 2 * w;  // this wave's two 16-row k tiles
+  for (int ki = 0; ki < 2; ++ki) {
+    const int ktok_base = (kt0 + ki) * 16;
+    if (ktok_base >= S) break;
+    int ktok = ktok_base + lm;
+    if (ktok >= S) ktok = S - 1;
+    f32x4 vacc[4], kacc[4];
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+      vacc[ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
+      kacc[ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      if (ks * 32 >= S) break;
+      // A fragments from the transposed global scratch (contiguous q)
+      const size_t abase = ((size_t)bh * S + ktok) * S + ks * 32 + lg * 8;
+      bf16x8 apd = *(const bf16x8*)(pdT + abase);
+      bf16x8 ads = *(const bf16x8*)(dsT + abase);
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        const int c = ct * 16 + lm;
+        bf16x8 bdo = *(const bf16x8*)lds_at(
+            lds, swz(16384 + c * 256 + (ks * 32 + lg * 8) * 2, c, 15));
+        vacc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(apd, bdo, vacc[ct], 0, 0, 0);
+        bf16x8 bq = *(const bf16x8*)lds_at(
+            lds, swz(0 + c * 256 + (ks * 32 + lg * 8) * 2, c, 15));
+        kacc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ads, bq, kacc[ct], 0, 0, 0);
+      }
+    }
+    ushort_t* dk = dqkv + (size_t)b * S * ts + (size_t)(h + hh) * ATT_D;
+    ushort_t* dv = dqkv + (size_t)b * S * ts + (size_t)(2 * h + hh) * ATT_D;
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = ktok_base + lg * 4 + r;
+        const int c = ct * 16 + lm;
+        if (row < S) {
+          dk[(size_t)row * ts + c] = f32_to_bf16(kacc[ct][r]);
+          dv[(size_t)row * ts + c] = f32_to_bf16(vacc[ct][r]);
+        }
+      }
+    }
+  }
+}
+
+SKY_EXPORT int sky_attn_bwd(uint64_t stream, uint64_t qkv, uint64_t dout,
+                            uint64_t mask, uint64_t m, uint64_t lsum,
+                            uint64_t pdT, uint64_t dsT, uint64_t dqkv,
+                            int64_t B, int64_t S, int64_t h, int64_t d,
+                            float scale, float keep, uint64_t salt,
+                            uint64_t state) {
+  if (d != ATT_D || S > ATT_SMAX) return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid((unsigned)(B * h));
+  bool hm = mask != 0;
+  if (hm)
+    hipLaunchKernelGGL((attn_bwd1_kernel<true>), grid, dim3(256), 80 * 1024, s,
+                       (const ushort_t*)qkv, (const ushort_t*)dout,
+                       (const ushort_t*)mask, (const float*)m,
+                       (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
+                       (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
+                       salt, (const unsigned long long*)state);
+  else
+    hipLaunchKernelGGL((attn_bwd1_kernel<false>), grid, dim3(256), 80 * 1024, s,
+                       (const ushort_t*)qkv, (const ushort_t*)dout,
+                       (const ushort_t*)mask, (const float*)m,
+                       (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
+                       (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
+                       salt, (const unsigned long long*)state);
+  hipLaunchKernelGGL(attn_bwd2_kernel, grid, dim3(256), 32 * 1024, s,
+                     (const ushort_t*)qkv, (const ushort_t*)dout,
+                     (const ushort_t*)pdT, (const ushort_t*)dsT,
+                     (ushort_t*)dqkv, (int)B, (int)S, (int)h);
+  LAUNCH_CHECK();
+  return 0;
+}
