@@ -37,8 +37,13 @@ def _build_plan(params, grads, masters, moms):
                 cnt, 0,
             ])
             off += cnt
+    # pinned staging so the H2D copy is legal inside hipGraph capture
     cpu = torch.tensor(rows, dtype=torch.int64)
-    return cpu.to(params[0].device), len(rows)
+    try:
+        cpu = cpu.pin_memory()
+    except RuntimeError:
+        pass
+    return cpu.to(params[0].device, non_blocking=True), len(rows)
 
 
 def multi_tensor_sgd(params, grads, masters, moms, lr, momentum, wd):

@@ -50,25 +50,28 @@ class GraphedTrainStep:
             optimizer.step()
             return loss
 
-        # warmup on a side stream (allocates grads, builds the SGD plan)
+        # warmup on a side stream (cudnn/hipblaslt heuristics, allocator)
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
             for _ in range(warmup_iters):
-                self._zero_grads(first=True)
+                self._drop_grads()
                 one_step()
         torch.cuda.current_stream().wait_stream(side)
 
-        self._grads = [p.grad for p in stage.parameters() if p.grad is not None]
+        # Capture with grads set to None: AccumulateGrad then STEALS each
+        # computed grad tensor (no zero-fill kernels, no += adds) and the
+        # graph pool reallocates the same addresses every replay. The SGD
+        # descriptor table is rebuilt inside capture (pinned H2D) once the
+        # grads exist.
+        self._drop_grads()
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
-            torch._foreach_zero_(self._grads)
             self.static_loss = one_step()
 
-    def _zero_grads(self, first: bool = False):
+    def _drop_grads(self):
         for p in self.stage.parameters():
-            if p.grad is not None:
-                p.grad.zero_()
+            p.grad = None
 
     def step(self, inputs, labels) -> float:
         for buf, t in zip(self.static_inputs, inputs):
