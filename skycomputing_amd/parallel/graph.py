@@ -61,13 +61,21 @@ class GraphedTrainStep:
 
         # Capture with grads set to None: AccumulateGrad then STEALS each
         # computed grad tensor (no zero-fill kernels, no += adds) and the
-        # graph pool reallocates the same addresses every replay. The SGD
-        # descriptor table is rebuilt inside capture (pinned H2D) once the
-        # grads exist.
+        # graph pool reallocates the same addresses every replay. Two
+        # graphs over one pool: g1 = tick+forward+loss+backward; the SGD
+        # descriptor is built BETWEEN captures (host allocs are illegal
+        # inside), then g2 = the fused SGD step.
         self._drop_grads()
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
-            self.static_loss = one_step()
+            F.rng_tick()
+            out = stage(*self.static_inputs)
+            self.static_loss = loss_fn(out, self.static_labels)
+            self.static_loss.backward()
+        optimizer.step()  # uncaptured: builds+caches the descriptor table
+        self.graph2 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph2, pool=self.graph.pool()):
+            optimizer.step()
 
     def _drop_grads(self):
         for p in self.stage.parameters():
@@ -79,6 +87,7 @@ class GraphedTrainStep:
                 buf.copy_(t, non_blocking=True)
         self.static_labels.copy_(labels, non_blocking=True)
         self.graph.replay()
+        self.graph2.replay()
         return self.static_loss
 
     def loss_value(self) -> float:
