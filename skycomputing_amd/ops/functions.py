@@ -85,20 +85,26 @@ class LayerNormFn(torch.autograd.Function):
         rows = x.numel() // cols
         dx = torch.empty_like(x)
         fast = cols % 8 == 0
-        alloc = torch.empty if fast else torch.zeros
-        dw32 = alloc(cols, dtype=torch.float32, device=x.device)
-        db32 = alloc(cols, dtype=torch.float32, device=x.device)
-        scratch = _red_scratch(cols, 2, x.device) if fast else None
+        if fast:
+            # the final reduction stage writes the param dtype directly
+            dw = torch.empty(cols, dtype=weight.dtype, device=x.device)
+            db = torch.empty(cols, dtype=weight.dtype, device=x.device)
+            scratch = _red_scratch(cols, 2, x.device)
+        else:
+            dw = torch.zeros(cols, dtype=torch.float32, device=x.device)
+            db = torch.zeros(cols, dtype=torch.float32, device=x.device)
+            scratch = None
         check(
             lib.sky_layernorm_bwd(
                 _stream(), ptr(dy), ptr(x), ptr(ctx.residual), ptr(weight),
-                ptr(mean), ptr(rstd), ptr(dx), ptr(dw32), ptr(db32),
+                ptr(mean), ptr(rstd), ptr(dx), ptr(dw), ptr(db),
                 ptr(scratch), rows, cols, _dt(x),
             ),
             "sky_layernorm_bwd",
         )
-        dw = dw32.to(weight.dtype)
-        db = db32.to(weight.dtype)
+        if not fast:
+            dw = dw.to(weight.dtype)
+            db = db.to(weight.dtype)
         return dx, dw, db, None, (dx if ctx.has_residual else None)
 
 
@@ -128,16 +134,17 @@ class BiasGeluFn(torch.autograd.Function):
         rows = x.numel() // cols
         dx = torch.empty_like(x)
         fast = cols % 8 == 0
-        db32 = (torch.empty if fast else torch.zeros)(cols, dtype=torch.float32, device=x.device)
+        db = (torch.empty(cols, dtype=bias.dtype, device=x.device) if fast
+              else torch.zeros(cols, dtype=torch.float32, device=x.device))
         scratch = _red_scratch(cols, 1, x.device) if fast else None
         check(
             lib.sky_bias_gelu_bwd(
-                _stream(), ptr(dy), ptr(x), ptr(bias), ptr(dx), ptr(db32),
+                _stream(), ptr(dy), ptr(x), ptr(bias), ptr(dx), ptr(db),
                 ptr(scratch), rows, cols, _dt(x)
             ),
             "sky_bias_gelu_bwd",
         )
-        return dx, db32.to(bias.dtype)
+        return dx, (db if fast else db.to(bias.dtype))
 
 
 class MaskedSoftmaxFn(torch.autograd.Function):
@@ -348,13 +355,17 @@ def colsum(src: torch.Tensor, out_dtype=None) -> torch.Tensor:
     src = src.contiguous()
     rows, cols = src.shape
     fast = cols % 8 == 0
-    out = (torch.empty if fast else torch.zeros)(cols, dtype=torch.float32, device=src.device)
+    want = out_dtype if (out_dtype is not None and fast) else torch.float32
+    out = (torch.empty if fast else torch.zeros)(cols, dtype=want, device=src.device)
     scratch = _red_scratch(cols, 1, src.device) if fast else None
     check(
-        lib.sky_colsum(_stream(), ptr(src), ptr(out), ptr(scratch), rows, cols, _dt(src)),
+        lib.sky_colsum(_stream(), ptr(src), ptr(out), ptr(scratch), rows, cols,
+                       _dt(src), _dt(out)),
         "sky_colsum",
     )
-    return out.to(out_dtype) if out_dtype is not None else out
+    if out_dtype is not None and out.dtype != out_dtype:
+        out = out.to(out_dtype)
+    return out
 
 
 class LinearBiasFn(torch.autograd.Function):

@@ -540,22 +540,29 @@ __global__ __launch_bounds__(256) void attn_bwd1_kernel(
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1) dot[r] += __shfl_xor(dot[r], off, 64);
     }
-    // dS = scale * P * (dP - dot); write dSrow (LDS) + dS^T, Pd^T (global)
+    // dS = scale * P * (dP - dot); write dSrow (LDS) + dS^T, Pd^T (global).
+    // The transposed stores pack this lane's 4 consecutive rows (lg*4+r)
+    // into ONE 8-byte store per (array, kt) — the store tail is
+    // issue-bound, not bandwidth-bound (guide T21).
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int row = qtok_base + lg * 4 + r;
+    for (int kt = 0; kt < 8; ++kt) {
+      if (kt >= NT) continue;
+      const int col = kt * 16 + lm;
+      ushort4_t ds4, pd4;
 #pragma unroll
-      for (int kt = 0; kt < 8; ++kt) {
-        if (kt >= NT) continue;
-        const int col = kt * 16 + lm;
+      for (int r = 0; r < 4; ++r) {
+        const int row = qtok_base + lg * 4 + r;
         const float ds = scale * pP[kt][r] * (pdP[kt][r] - dot[r]);
         const int rl = qi * 16 + lg * 4 + r;
         *(ushort_t*)lds_at(lds, swz(49152 + w * 8192 + rl * 256 + col * 2, rl, 15)) =
             f32_to_bf16(ds);
-        if (row < S && col < S) {
-          dsT[((size_t)bh * S + col) * S + row] = f32_to_bf16(ds);
-          pdT[((size_t)bh * S + col) * S + row] = f32_to_bf16(pPd[kt][r]);
-        }
+        ds4[r] = f32_to_bf16((row < S) ? ds : 0.f);
+        pd4[r] = f32_to_bf16((row < S) ? pPd[kt][r] : 0.f);
+      }
+      if (col < S) {
+        const size_t o = ((size_t)bh * S + col) * S + qtok_base + lg * 4;
+        *(ushort4_t*)(dsT + o) = ds4;
+        *(ushort4_t*)(pdT + o) = pd4;
       }
     }
     // dQ = dS K : A = dSrow (this wave's LDS), B = Kt

@@ -126,9 +126,9 @@ __global__ __launch_bounds__(BLOCK) void colsum_part_kernel(
 
 // two-level: block covers 16 columns x 16 slab-groups (each thread sums
 // nslabs/16 slabs of one column), LDS tree, then one write per column.
-template <int BLOCK>
+template <int BLOCK, int DTOUT>
 __global__ __launch_bounds__(BLOCK) void colsum_final_kernel(
-    const float* __restrict__ scratch, float* __restrict__ out, int64_t cols,
+    const float* __restrict__ scratch, void* __restrict__ out, int64_t cols,
     int64_t nslabs) {
   __shared__ float lds[16][17];
   const int c = threadIdx.x & 15;
@@ -147,7 +147,7 @@ __global__ __launch_bounds__(BLOCK) void colsum_final_kernel(
     float t = 0.f;
 #pragma unroll
     for (int j = 0; j < 16; ++j) t += lds[j][c];
-    out[col] = t;
+    store_elem<DTOUT>(out, col, t);
   }
 }
 
@@ -168,8 +168,9 @@ __global__ __launch_bounds__(BLOCK) void colsum_kernel(
 // (cols % 8 == 0); pass scratch = nullptr to force the atomic fallback
 // (out must then be pre-zeroed).
 template <int DT>
-static void launch_colsum(hipStream_t s, const void* src, float* out,
-                          float* scratch, int64_t rows, int64_t cols) {
+static void launch_colsum(hipStream_t s, const void* src, void* out,
+                          float* scratch, int64_t rows, int64_t cols,
+                          int dtout) {
   if (cols % 8 == 0 && scratch != nullptr) {
     constexpr int BLOCK = 128;
     const int64_t cols8 = cols / 8;
@@ -179,8 +180,12 @@ static void launch_colsum(hipStream_t s, const void* src, float* out,
     hipLaunchKernelGGL((colsum_part_kernel<DT, BLOCK>), grid, dim3(BLOCK), 0,
                        s, src, scratch, rows, cols8, slab);
     dim3 g2((unsigned)((cols + 15) / 16));
-    hipLaunchKernelGGL((colsum_final_kernel<256>), g2, dim3(256), 0, s,
-                       scratch, out, cols, nslabs);
+    if (dtout == DT_BF16)
+      hipLaunchKernelGGL((colsum_final_kernel<256, DT_BF16>), g2, dim3(256), 0,
+                         s, scratch, out, cols, nslabs);
+    else
+      hipLaunchKernelGGL((colsum_final_kernel<256, DT_F32>), g2, dim3(256), 0,
+                         s, scratch, out, cols, nslabs);
     return;
   }
   constexpr int BLOCK = 256;
@@ -189,18 +194,19 @@ static void launch_colsum(hipStream_t s, const void* src, float* out,
     slab *= 2;
   dim3 grid((unsigned)((cols + BLOCK - 1) / BLOCK),
             (unsigned)((rows + slab - 1) / slab));
+  (void)dtout;  // atomic fallback accumulates fp32 into `out`
   hipLaunchKernelGGL((colsum_kernel<DT, BLOCK>), grid, dim3(BLOCK), 0, s, src,
-                     out, rows, cols, slab);
+                     (float*)out, rows, cols, slab);
 }
 
 SKY_EXPORT int sky_colsum(uint64_t stream, uint64_t src, uint64_t out,
                           uint64_t scratch, int64_t rows, int64_t cols,
-                          int dt) {
+                          int dt, int dtout) {
   hipStream_t s = (hipStream_t)stream;
   if (dt == DT_F32)
-    launch_colsum<DT_F32>(s, (const void*)src, (float*)out, (float*)scratch, rows, cols);
+    launch_colsum<DT_F32>(s, (const void*)src, (void*)out, (float*)scratch, rows, cols, dtout);
   else
-    launch_colsum<DT_BF16>(s, (const void*)src, (float*)out, (float*)scratch, rows, cols);
+    launch_colsum<DT_BF16>(s, (const void*)src, (void*)out, (float*)scratch, rows, cols, dtout);
   LAUNCH_CHECK();
   return 0;
 }
@@ -232,9 +238,9 @@ SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
                          (const void*)dy, (const void*)x, (const void*)b, (void*)dx, n, cols);
   }
   if (dt == DT_F32)
-    launch_colsum<DT_F32>(s, (const void*)dx, (float*)db, (float*)scratch, rows, cols);
+    launch_colsum<DT_F32>(s, (const void*)dx, (void*)db, (float*)scratch, rows, cols, dt);
   else
-    launch_colsum<DT_BF16>(s, (const void*)dx, (float*)db, (float*)scratch, rows, cols);
+    launch_colsum<DT_BF16>(s, (const void*)dx, (void*)db, (float*)scratch, rows, cols, dt);
   LAUNCH_CHECK();
   return 0;
 }

@@ -246,10 +246,10 @@ __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_part_kernel(
 
 // two-level: block covers 16 columns x 16 slab-groups, LDS tree, one
 // write per column (see colsum_final_kernel).
-template <int BLOCK>
+template <int BLOCK, int DTOUT>
 __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_final_kernel(
-    const float* __restrict__ scratch, float* __restrict__ dw,
-    float* __restrict__ db, int64_t cols, int64_t nslabs) {
+    const float* __restrict__ scratch, void* __restrict__ dw,
+    void* __restrict__ db, int64_t cols, int64_t nslabs) {
   __shared__ float ldsw[16][17];
   __shared__ float ldsb[16][17];
   const int c = threadIdx.x & 15;
@@ -271,8 +271,8 @@ __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_final_kernel(
     float tw = 0.f, tb = 0.f;
 #pragma unroll
     for (int j = 0; j < 16; ++j) { tw += ldsw[j][c]; tb += ldsb[j][c]; }
-    dw[col] = tw;
-    db[col] = tb;
+    store_elem<DTOUT>(dw, col, tw);
+    store_elem<DTOUT>(db, col, tb);
   }
 }
 
@@ -381,9 +381,14 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
     else              { if (has_res) LNWBP(DT_BF16, true); else LNWBP(DT_BF16, false); }
 #undef LNWBP
     dim3 g2((unsigned)((cols + 15) / 16));
-    hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256>), g2, dim3(256), 0, s,
-                       (const float*)scratch, (float*)dw, (float*)db, cols,
-                       nslabs);
+    if (dt == DT_BF16)
+      hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256, DT_BF16>), g2, dim3(256),
+                         0, s, (const float*)scratch, (void*)dw, (void*)db,
+                         cols, nslabs);
+    else
+      hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256, DT_F32>), g2, dim3(256),
+                         0, s, (const float*)scratch, (void*)dw, (void*)db,
+                         cols, nslabs);
   } else {
     constexpr int BLOCK = 256;
     // slab sized so the grid fills 256 CUs (cols/256 col-blocks * row-slabs)

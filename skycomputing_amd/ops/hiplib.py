@@ -54,7 +54,7 @@ def _register_signatures(lib):
         #                   strm  x    y    n   keep salt state dt
         "sky_dropout_bwd": [u64, u64, u64, i64, f32, u64, u64, i32],
         "sky_rng_tick": [u64, u64],  # strm, state ptr
-        "sky_colsum": [u64, u64, u64, u64, i64, i64, i32],  # strm src out32 scratch rows cols dt
+        "sky_colsum": [u64, u64, u64, u64, i64, i64, i32, i32],  # strm src out scratch rows cols dt dtout
         "sky_mfma_probe": [u64, u64, u64, u64],  # strm A B D
         "sky_glds_probe": [u64, u64, u64, i32],
         "sky_gemm": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, i64, i64, i32, i32, i32, i32, i32],
