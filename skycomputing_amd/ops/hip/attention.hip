@@ -33,11 +33,14 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define ATT_SMAX 128
 #define ATT_D 64
 
-// LDS layout (bytes): K [128][64]bf16 at 0 (16K), Vt [64][128]bf16 at 16K,
-// P per wave [32][128]bf16 at 32K + w*8K. Total 64K.
+// LDS layout (bytes): K [128][64]bf16 at 0 (16K); P per wave [32][128]
+// at 16K + w*8K (32K). After QK^T, V^T [64][128] ALIASES K's region
+// (K is dead) — total 48K, so 3 workgroups fit per CU (160K LDS) instead
+// of 2, +50% latency-hiding concurrency.
 #define K_OFF 0
-#define VT_OFF (16 * 1024)
-#define P_OFF (32 * 1024)
+#define VT_OFF 0
+#define P_OFF (16 * 1024)
+#define ATT_LDS_BYTES (48 * 1024)
 
 DEV void* lds_at(char* base, int byte) { return (void*)(base + byte); }
 
@@ -72,11 +75,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   // ---- zero LDS when S < full tile coverage (avoid NaN poisoning) ----
   if (S < ATT_SMAX) {
-    for (int u = tid; u < (64 * 1024) / 16; u += 256)
+    for (int u = tid; u < ATT_LDS_BYTES / 16; u += 256)
       *(ushort8_t*)lds_at(lds, u * 16) = (ushort8_t)(ushort_t)0;
     __syncthreads();
-  } else {
-    // P tiles still need zeroing for partial last q tile? S==128 -> full.
   }
 
   // ---- stage K into LDS [S][64] via global_load_lds: the XOR swizzle
@@ -88,17 +89,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     __builtin_amdgcn_global_load_lds(
         (att_gas)(kbase + (size_t)tok * ts + c16s * 8),
         (att_las)lds_at(lds, K_OFF + u * 16), 16, 0, 0);
-  }
-  // ---- stage V transposed into LDS [64][S] ----
-  for (int u = tid; u < S * 8; u += 256) {
-    const int tok = u >> 3;
-    const int c16 = u & 7;
-    ushort8_t v = *(const ushort8_t*)(vbase + (size_t)tok * ts + c16 * 8);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int c = c16 * 8 + j;
-      *(ushort_t*)lds_at(lds, swz(VT_OFF + c * 256 + tok * 2, c, 15)) = v[j];
-    }
   }
   // ---- per-wave constants (issued BEFORE the barrier so the global
   // loads land under the staging latency) ----
@@ -257,6 +247,25 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   if (PROBS_MODE) return;
 
+  // ---- stage V transposed over K's (now dead) region ----
+  __syncthreads();  // all waves done reading K and writing their P tiles
+  if (S < ATT_SMAX) {
+    for (int u = tid; u < (16 * 1024) / 16; u += 256)
+      *(ushort8_t*)lds_at(lds, VT_OFF + u * 16) = (ushort8_t)(ushort_t)0;
+    __syncthreads();  // zero before scattered staging writes land
+  }
+  for (int u = tid; u < S * 8; u += 256) {
+    const int tok = u >> 3;
+    const int c16 = u & 7;
+    ushort8_t v = *(const ushort8_t*)(vbase + (size_t)tok * ts + c16 * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c16 * 8 + j;
+      *(ushort_t*)lds_at(lds, swz(VT_OFF + c * 256 + tok * 2, c, 15)) = v[j];
+    }
+  }
+  __syncthreads();
+
   // ---- PV: O[32 rows][64] per wave ----
   // (same-wave LDS write->read; compiler inserts the lgkm waits)
   for (int qi = 0; qi < 2; ++qi) {
@@ -350,7 +359,7 @@ static int attn_launch(uint64_t stream, uint64_t qkv, uint64_t mask,
   if (d != ATT_D || S > ATT_SMAX) return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
   dim3 grid((unsigned)(B * h));
-  size_t lds_bytes = 64 * 1024;
+  size_t lds_bytes = ATT_LDS_BYTES;
   bool hm = mask != 0;
 #define ATT(HM, SML, PM)                                                      \
   hipLaunchKernelGGL((attn_fwd_kernel<HM, SML, PM>), grid, dim3(256),         \
