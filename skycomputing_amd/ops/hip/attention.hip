@@ -49,7 +49,7 @@ DEV int swz(int byte_addr, int row, int rmask) {
 }
 
 template <bool HAS_MASK, bool SAVE_ML, bool PROBS_MODE>
-__global__ __launch_bounds__(256) void attn_fwd_kernel(
+__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ mask,
     ushort_t* __restrict__ out, float* __restrict__ m_io,
     float* __restrict__ l_io, ushort_t* __restrict__ p_out,
@@ -123,6 +123,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
     for (int dv = 0; dv < 4; ++dv) oacc[qi][dv] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
+#pragma clang loop unroll(disable)
   for (int qi = 0; qi < 2; ++qi) {
     const int qtile = qt0 + qi;
     const int qtok_base = qtile * 16;
@@ -143,8 +144,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         sacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, sacc[kt], 0, 0, 0);
       }
     }
-    // softmax over this wave's 4 rows per (lg, r)
-    float z[8][4];
+    // softmax over this wave's 4 rows per (lg, r); the transformed
+    // scores overwrite sacc in place (register budget)
     float mx[4], sm[4], inv[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) mx[r] = -3.0e38f;
@@ -154,7 +155,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float zz = sacc[kt][r] * scale + mval[kt];
-        z[kt][r] = zz;
+        sacc[kt][r] = zz;
         mx[r] = fmaxf(mx[r], zz);
       }
     }
@@ -185,8 +186,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       if (kt >= NT) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float p = __expf(z[kt][r] - (PROBS_MODE ? mrow[r] : mx[r]));
-        z[kt][r] = p;
+        float p = __expf(sacc[kt][r] - (PROBS_MODE ? mrow[r] : mx[r]));
+        sacc[kt][r] = p;
         sm[r] += p;
       }
     }
@@ -223,7 +224,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int kt = 0; kt < 8; ++kt) {
         if (kt >= NT) continue;
         const int col = kt * 16 + lm;
-        float p = z[kt][r] * inv[r];
+        float p = sacc[kt][r] * inv[r];
         float pd = p;
         if (keep < 1.f) {
           const uint64_t zz = kt < 4 ? z0 : z1;
@@ -410,7 +411,7 @@ SKY_EXPORT int sky_attn_probs(uint64_t stream, uint64_t qkv, uint64_t mask,
 // ============================================================================
 
 template <bool HAS_MASK>
-__global__ __launch_bounds__(256) void attn_bwd1_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd1_kernel(
     const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ dout,
     const ushort_t* __restrict__ mask, const float* __restrict__ m_io,
     const float* __restrict__ l_io, ushort_t* __restrict__ pdT,
@@ -482,6 +483,7 @@ __global__ __launch_bounds__(256) void attn_bwd1_kernel(
   }
   __syncthreads();
 
+#pragma clang loop unroll(disable)
   for (int qi = 0; qi < 2; ++qi) {
     const int qtok_base = (qt0 + qi) * 16;
     if (qtok_base >= S) break;
