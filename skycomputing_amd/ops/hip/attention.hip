@@ -507,8 +507,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd1_kernel(
         dacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado_all[qi][ks], bv, dacc[kt], 0, 0, 0);
       }
     }
-    // P, Pd, dP, rowdot, dS  (rows r of this lane: qtok_base + lg*4 + r)
+    // P, Pd, dP, rowdot, dS. Register budget: p/pd/dp are NOT kept in
+    // arrays (3x32 VGPRs would spill); the store loop recomputes them
+    // from the still-live sacc/dacc and the cached dropout hashes.
     float mrow[4], lrow[4], dot[4];
+    uint64_t zs[4][2];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = qtok_base + lg * 4 + r;
@@ -516,33 +519,28 @@ __global__ __launch_bounds__(256, 2) void attn_bwd1_kernel(
       mrow[r] = m_io[(size_t)bh * S + rr];
       lrow[r] = 1.f / l_io[(size_t)bh * S + rr];
       dot[r] = 0.f;
+      zs[r][0] = zs[r][1] = 0;
+      if (keep < 1.f) {
+        const uint64_t base = ((uint64_t)bh * S + row) * 16 + lm;
+        zs[r][0] = rng_hash(seed, base * 2);
+        zs[r][1] = rng_hash(seed, base * 2 + 1);
+      }
     }
-    float pP[8][4], pPd[8][4], pdP[8][4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = qtok_base + lg * 4 + r;
-      uint64_t z0 = 0, z1 = 0;
-      if (keep < 1.f) {
-        const uint64_t base = ((uint64_t)bh * S + row) * 16 + lm;
-        z0 = rng_hash(seed, base * 2);
-        z1 = rng_hash(seed, base * 2 + 1);
-      }
 #pragma unroll
       for (int kt = 0; kt < 8; ++kt) {
         if (kt >= NT) continue;
-        float p = __expf(sacc[kt][r] * scale + mval[kt] - mrow[r]) * lrow[r];
         const int col = kt * 16 + lm;
+        float p = __expf(sacc[kt][r] * scale + mval[kt] - mrow[r]) * lrow[r];
         if (!(row < S && col < S)) p = 0.f;
-        float pd = p, dp = dacc[kt][r];
+        float dp = dacc[kt][r];
         if (keep < 1.f) {
-          const uint64_t zz = kt < 4 ? z0 : z1;
+          const uint64_t zz = kt < 4 ? zs[r][0] : zs[r][1];
           bool kbit = (unsigned)((zz >> (16 * (kt & 3))) & 0xFFFFu) < keep16;
-          pd = kbit ? p * inv_keep : 0.f;
           dp = kbit ? dp * inv_keep : 0.f;
         }
-        pP[kt][r] = p;
-        pPd[kt][r] = pd;
-        pdP[kt][r] = dp;
         dot[r] += dp * p;
       }
     }
@@ -551,10 +549,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd1_kernel(
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1) dot[r] += __shfl_xor(dot[r], off, 64);
     }
-    // dS = scale * P * (dP - dot); write dSrow (LDS) + dS^T, Pd^T (global).
-    // The transposed stores pack this lane's 4 consecutive rows (lg*4+r)
-    // into ONE 8-byte store per (array, kt) — the store tail is
-    // issue-bound, not bandwidth-bound (guide T21).
+    // dS = scale * P * (dP - dot); write dSrow (LDS) + dS^T, Pd^T (global,
+    // 4 consecutive rows packed per 8-byte store — issue-bound tail, T21)
 #pragma unroll
     for (int kt = 0; kt < 8; ++kt) {
       if (kt >= NT) continue;
@@ -563,12 +559,21 @@ __global__ __launch_bounds__(256, 2) void attn_bwd1_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = qtok_base + lg * 4 + r;
-        const float ds = scale * pP[kt][r] * (pdP[kt][r] - dot[r]);
+        float p = __expf(sacc[kt][r] * scale + mval[kt] - mrow[r]) * lrow[r];
+        if (!(row < S && col < S)) p = 0.f;
+        float pd = p, dp = dacc[kt][r];
+        if (keep < 1.f) {
+          const uint64_t zz = kt < 4 ? zs[r][0] : zs[r][1];
+          bool kbit = (unsigned)((zz >> (16 * (kt & 3))) & 0xFFFFu) < keep16;
+          pd = kbit ? p * inv_keep : 0.f;
+          dp = kbit ? dp * inv_keep : 0.f;
+        }
+        const float ds = scale * p * (dp - dot[r]);
         const int rl = qi * 16 + lg * 4 + r;
         *(ushort_t*)lds_at(lds, swz(49152 + w * 8192 + rl * 256 + col * 2, rl, 15)) =
             f32_to_bf16(ds);
-        ds4[r] = f32_to_bf16((row < S) ? ds : 0.f);
-        pd4[r] = f32_to_bf16((row < S) ? pPd[kt][r] : 0.f);
+        ds4[r] = f32_to_bf16(ds);
+        pd4[r] = f32_to_bf16(pd);
       }
       if (col < S) {
         const size_t o = ((size_t)bh * S + col) * S + qtok_base + lg * 4;

@@ -57,3 +57,19 @@ k = qkv[:, :, 1].permute(0, 2, 1, 3).contiguous()
 v = qkv[:, :, 2].permute(0, 2, 1, 3).contiguous()
 print(f"torch sdpa (no drop): "
       f"{bench(lambda: torch.nn.functional.scaled_dot_product_attention(q, k, v)):8.1f} us")
+
+# fused backward timing
+dout = torch.randn(B, S, h, d, dtype=torch.bfloat16, device="cuda")
+pdT = torch.empty(B, h, S, S, dtype=torch.bfloat16, device="cuda")
+dsT = torch.empty_like(pdT)
+dqkv = torch.empty_like(qkv)
+
+
+def run_bwd(keep=0.9):
+    check(lib.sky_attn_bwd(stream, ptr(qkv), ptr(dout), ptr(mask), ptr(m),
+                           ptr(l), ptr(pdT), ptr(dsT), ptr(dqkv),
+                           B, S, h, d, scale, keep, 123, 0), "bwd")
+
+
+print(f"attn_bwd  keep=0.9: {bench(lambda: run_bwd(0.9)):8.1f} us")
+print(f"attn_bwd  keep=1.0: {bench(lambda: run_bwd(1.0)):8.1f} us")
