@@ -1,0 +1,153 @@
+"""Unit tests for the smaller subsystems: workers, stimulator, timers,
+datasets, optimizer, parameter server."""
+
+from __future__ import annotations
+
+import os
+
+import pytest
+import torch
+
+from skycomputing_amd.dataset import (
+    GlueDataset, RandomMlpDataset, RandomTokenGenerator, SyntheticGlueDataset,
+)
+from skycomputing_amd.dynamics import ParameterServer, Worker, WorkerManager
+from skycomputing_amd.optim import FusedSGD
+from skycomputing_amd.stimulator import Stimulator
+from skycomputing_amd.timer import DeviceTimer, DistributedTimer
+
+from .helpers import tiny_bert_cfg
+
+
+def test_worker_roundtrip_and_manager():
+    w = Worker(rank=2, extra_config=dict(slowdown=1.5))
+    w.model_config = (3, 7)
+    w.order = 0
+    w2 = Worker.from_dict(w.to_dict())
+    assert w2.rank == 2 and w2.model_config == (3, 7) and w2.extra_config["slowdown"] == 1.5
+
+    wm = WorkerManager.from_world(3)
+    wm.assign_model_to_worker(0, (4, 8), order=1)
+    wm.assign_model_to_worker(1, (0, 4), order=0)
+    wm.assign_model_to_worker(2, (8, 10), order=2)
+    order = [w.rank for w in wm.pipeline_order()]
+    assert order == [1, 0, 2]
+    wm.remove_worker(1)
+    assert len(wm) == 2
+    with pytest.raises(KeyError):
+        wm.get_worker_by_rank(1)
+    with pytest.raises(ValueError):
+        wm.add_worker(Worker(rank=0))
+
+
+def test_stimulator_deterministic_ranges():
+    s1, s2 = Stimulator(8, seed=42), Stimulator(8, seed=42)
+    assert (s1.compute_factors == s2.compute_factors).all()
+    assert all(1 <= f < 4 for f in s1.compute_factors)
+    assert all(1 <= f < 3 for f in s1.memory_factors)
+    assert all(1 <= f < 2 for f in s1.network_factors)
+    res = {0: {"time": 1.0, "avai_mem": 100.0}}
+    scaled = s1.scale_benchmark(res)
+    assert scaled[0]["time"] == pytest.approx(s1.compute_factor(0))
+    assert scaled[0]["avai_mem"] == pytest.approx(100.0 / s1.memory_factor(0))
+
+
+def test_device_timer_cpu():
+    import time
+
+    t = DeviceTimer(use_cuda=False)
+    t.start(); time.sleep(0.02); t.stop()
+    t.start(); time.sleep(0.01); t.stop()
+    assert 0.025 < t.elapsed() < 0.5
+    assert 0.008 < t.last() < 0.3
+    t.reset()
+    assert t.elapsed() == 0.0
+
+
+def test_distributed_timer_intervals():
+    import time
+
+    dt = DistributedTimer()
+    dt.add_timestamp("x"); time.sleep(0.01); dt.add_timestamp("x")
+    assert dt.get_prev_interval("x") > 0.005
+    assert dt.get_prev_interval("missing") == 0.0
+    dt.clean()
+    assert dt.get_prev_interval("x") == 0.0
+
+
+def test_synthetic_glue_schema_and_determinism():
+    a = SyntheticGlueDataset(size=16, max_seq_length=8, vocab_size=100, seed=5)
+    b = SyntheticGlueDataset(size=16, max_seq_length=8, vocab_size=100, seed=5)
+    (ids, mask, tids), label = a[3]
+    assert ids.shape == (8,) and mask.shape == (8,) and tids.shape == (8,)
+    assert torch.equal(a[3][0][0], b[3][0][0])
+    assert (a.input_ids < 100).all()
+    # type ids only where attended
+    assert ((a.token_type_ids == 1) <= (a.attention_mask == 1)).all()
+
+
+def test_glue_dataset_tokenizes_from_disk(tmp_path):
+    vocab = ["[PAD]", "[UNK]", "[CLS]", "[SEP]", "[MASK]",
+             "the", "cat", "sat", "mat", "dog", "ran"]
+    (tmp_path / "vocab.txt").write_text("\n".join(vocab) + "\n")
+    rows = ["index\tsentence1\tsentence2\tgold_label",
+            "0\tthe cat sat\tthe dog ran\tentailment",
+            "1\tthe mat\tthe cat\tneutral"]
+    (tmp_path / "train.tsv").write_text("\n".join(rows) + "\n")
+    try:
+        ds = GlueDataset(str(tmp_path), task="mnli", max_seq_length=16)
+    except ImportError:
+        pytest.skip("transformers unavailable")
+    assert len(ds) == 2
+    (ids, mask, tids), label = ds[0]
+    assert ids.shape == (16,)
+    assert ids[0] == 2  # [CLS]
+    assert int(label) == 1  # entailment
+    assert mask.sum() >= 7
+
+
+def test_random_token_generator():
+    g = RandomTokenGenerator(batch_size=2, seq_len=4, vocab_size=10)
+    ids, mask, tids = g.generate()
+    assert ids.shape == (2, 4) and (ids < 10).all()
+
+
+def test_fused_sgd_state_dict_roundtrip():
+    p = [torch.randn(8, requires_grad=True)]
+    opt = FusedSGD(p, lr=0.1, momentum=0.9, master_weights=True)
+    p[0].grad = torch.randn(8)
+    opt.step()
+    sd = opt.state_dict()
+    opt2 = FusedSGD([p[0]], lr=0.1, momentum=0.9, master_weights=True)
+    opt2.load_state_dict(sd)
+    assert torch.allclose(opt2.masters[0], opt.masters[0])
+    assert torch.allclose(opt2.momentum_bufs[0], opt.momentum_bufs[0])
+
+
+def test_parameter_server_save_load(tmp_path):
+    ps = ParameterServer(2)
+    ps.update_weights({"w": torch.randn(3)}, 0)
+    ps.update_weights({"w": torch.randn(3)}, 1)
+    path = str(tmp_path / "epoch_1.pth")
+    ps.save_weights_to_file(path, meta={"epoch": 1})
+    ps2 = ParameterServer(2)
+    meta = ps2.load_weights_from_file(path)
+    assert meta["epoch"] == 1
+    assert torch.allclose(ps2.get_state_dict(0)["w"], ps.get_state_dict(0)["w"])
+    ps3 = ParameterServer(3)
+    with pytest.raises(ValueError):
+        ps3.load_weights_from_file(path)
+    ps4 = ParameterServer(2)
+    with pytest.raises(RuntimeError):
+        ps4.save_weights_to_file(str(tmp_path / "x.pth"))
+
+
+def test_config_registry_hooks_buildable():
+    from skycomputing_amd.builder import build_hook
+    from skycomputing_amd.runner import CheckpointHook, StopHook, TimerHook
+
+    assert isinstance(build_hook(dict(layer_type="TimerHook")), TimerHook)
+    assert isinstance(build_hook(dict(layer_type="StopHook", root=".")), StopHook)
+    assert isinstance(
+        build_hook(dict(layer_type="CheckpointHook", save_path="/tmp/x")), CheckpointHook
+    )
