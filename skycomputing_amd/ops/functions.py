@@ -33,7 +33,7 @@ def _stream() -> int:
     return torch.cuda.current_stream().cuda_stream
 
 
-_CS_SLABS = 128
+_CS_SLABS = 256
 
 
 def _red_scratch(cols: int, pairs: int, device) -> torch.Tensor:

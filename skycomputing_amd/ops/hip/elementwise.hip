@@ -102,7 +102,7 @@ __global__ __launch_bounds__(256) void bias_gelu_bwd_dx_kernel(
 // no zero-init: stage 1 writes per-slab partials [nslabs][cols] fp32 to a
 // scratch buffer (thread owns 8 consecutive columns, 16 B loads); stage 2
 // reduces the slab axis and WRITES the result.
-#define CS_SLABS 128
+#define CS_SLABS 256
 
 template <int DT, int BLOCK>
 __global__ __launch_bounds__(BLOCK) void colsum_part_kernel(

@@ -205,7 +205,7 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_vec_kernel(
 // column-parallel dw/db, two-stage, no atomics, outputs need no zero-init:
 // stage 1 writes per-slab partials to scratch [nslabs][2*cols] fp32
 // (dw partial at [y][c], db partial at [y][cols+c]); stage 2 reduces.
-#define LN_SLABS 128
+#define LN_SLABS 256
 
 template <int DT, bool HAS_RES, int BLOCK>
 __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_part_kernel(
