@@ -14,7 +14,7 @@ import torch
 from . import hiplib
 from .hiplib import check
 
-_SLAB = 1 << 20  # elements per descriptor
+_SLAB = 1 << 18  # elements per descriptor (smaller slabs -> more blocks, shorter serial chains)
 
 _PLAN_CACHE: dict[int, tuple] = {}
 
