@@ -34,7 +34,7 @@ def parse_args():
     p.add_argument("--seq", type=int, default=128)
     p.add_argument("--allocate", default="even", choices=["even", "dynamic", "optimal"])
     p.add_argument("--microbatches", type=int, default=0)
-    p.add_argument("--schedule", default="gpipe", choices=["gpipe", "sequential"])
+    p.add_argument("--schedule", default="gpipe", choices=["gpipe", "sequential", "1f1b"])
     p.add_argument("--slowdowns", default="")
     p.add_argument("--stimulate", action="store_true")
     p.add_argument("--dropout", type=float, default=0.1)

@@ -65,7 +65,7 @@ train_config = dict(
     max_iter=30,
     optimizer=dict(lr=1e-3, momentum=0.0, weight_decay=0.0),
     num_microbatches=int(os.environ.get("SKY_MICROBATCHES", 0)),  # 0 = auto
-    schedule="gpipe",
+    schedule=os.environ.get("SKY_SCHEDULE", "gpipe"),  # gpipe | 1f1b | sequential
     log_interval=1,
     hooks=[
         dict(layer_type="TimerHook"),

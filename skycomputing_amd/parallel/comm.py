@@ -103,6 +103,38 @@ class CommContext:
             out.append(buf)
         return out
 
+    def cached_recv_meta(self, src: int, key: str):
+        return self._meta_cache_recv.get((src, key))
+
+    def fused_send_recv(self, sends, recvs):
+        """Post sends and recvs as ONE batch_isend_irecv — required when
+        traffic crosses in both directions between a rank pair (1F1B steady
+        state), where sequential blocking rendezvous ops can deadlock.
+
+        sends: [(tensor_list, dst)], recvs: [(spec_list, src)] with specs
+        (shape, dtype, requires_grad). Returns one buffer list per recv.
+        """
+        ops = []
+        for tensors, dst in sends:
+            for t in tensors:
+                ops.append(dist.P2POp(dist.isend, t.detach().contiguous(), dst))
+        out = []
+        for specs, src in recvs:
+            bufs = []
+            for shape, dtype, req in specs:
+                b = torch.empty(shape, dtype=dtype, device=self.device)
+                bufs.append(b)
+                ops.append(dist.P2POp(dist.irecv, b, src))
+            out.append(bufs)
+        if ops:
+            for r in dist.batch_isend_irecv(ops):
+                r.wait()
+        for (specs, _src), bufs in zip(recvs, out):
+            for (shape, dtype, req), b in zip(specs, bufs):
+                if req:
+                    b.requires_grad_(True)
+        return out
+
     def reset_channels(self):
         """Drop cached channel metadata (call after re-allocation changes
         the partition and therefore the boundary payloads).

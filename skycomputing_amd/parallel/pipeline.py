@@ -17,6 +17,10 @@ Schedules:
     deliberate capability extension over the reference (which has no
     microbatching) — blocking P2P is deadlock-free here
     because no fwd and bwd hop between a rank pair can cross.
+  * ``1f1b``     — one-forward-one-backward: same bubble as GPipe but peak
+    activation memory ~ num_stages microbatches instead of M. Steady-state
+    traffic crosses in both directions between neighbors, so the crossing
+    pairs are posted as single batch_isend_irecv calls.
 
 Gradient scaling: each microbatch loss is divided by M, so gradients equal
 the full-batch gradient (mean-reduction losses).
@@ -193,17 +197,124 @@ class PipelineEngine:
         mb_inputs = self._split(inputs, M) if self.is_first else [None] * M
         mb_labels = self._split(labels, M) if self.is_last else [None] * M
 
-        saved = []
-        for m in range(M):
-            saved.append(self._forward_microbatch(mb_inputs[m]))
-        total = 0.0
-        for m in range(M):
-            held, out = saved[m]
-            loss = self._backward_microbatch(held, out, mb_labels[m], M)
-            if loss is not None:
-                total += float(loss.detach())
-        saved.clear()
+        if schedule == "1f1b" and M > 1 and self.plan.num_stages > 1:
+            total = self._run_1f1b(mb_inputs, mb_labels, M)
+        else:
+            saved = []
+            for m in range(M):
+                saved.append(self._forward_microbatch(mb_inputs[m]))
+            total = 0.0
+            for m in range(M):
+                held, out = saved[m]
+                loss = self._backward_microbatch(held, out, mb_labels[m], M)
+                if loss is not None:
+                    total += float(loss.detach())
+            saved.clear()
         return self._broadcast_loss(total / M if self.is_last else None)
+
+    # ---------------- 1F1B ----------------
+    # One-forward-one-backward: peak activation memory ~ num_stages
+    # microbatches instead of M. Steady-state traffic crosses in both
+    # directions between neighbor ranks, so the two crossing pairs
+    # (send-fwd + recv-bwd with next; send-bwd + recv-fwd with prev) are
+    # posted as single batched P2P calls (comm.fused_send_recv).
+
+    def _grad_specs(self, out):
+        return [
+            (tuple(t.shape), t.dtype, False)
+            for t in out
+            if torch.is_tensor(t) and t.requires_grad
+        ]
+
+    def _run_1f1b(self, mb_inputs, mb_labels, M: int) -> float:
+        from collections import deque
+
+        N = self.plan.num_stages
+        s = self.stage_idx
+        warmup = min(N - 1 - s, M)
+        saved = deque()
+        total = 0.0
+        fwd_i = 0
+        bwd_i = 0
+
+        # ---- warmup: forward-only phase (plain blocking P2P is safe:
+        # all traffic flows downstream) ----
+        for _ in range(warmup):
+            saved.append(self._forward_microbatch(mb_inputs[fwd_i]))
+            fwd_i += 1
+
+        pending_input = None  # fwd input received by a fused call
+        remaining = M - warmup
+        for i in range(remaining):
+            # forward one microbatch
+            if self.is_first:
+                held, out = (), _to_tuple(self.stage(*_to_tuple(mb_inputs[fwd_i])))
+            else:
+                if pending_input is None:
+                    args = tuple(self.comm.recv_tensors(self.prev_rank, "fwd"))
+                else:
+                    args = tuple(pending_input)
+                    pending_input = None
+                held, out = args, _to_tuple(self.stage(*args))
+            fwd_i += 1
+            saved.append((held, out))
+
+            # send fwd output + receive bwd grads (fused with next rank)
+            held_b, out_b = saved.popleft()
+            if self.is_last:
+                logits = out_b[0] if len(out_b) == 1 else out_b
+                loss = self.loss_fn(logits, mb_labels[bwd_i].to(logits.device))
+                (loss / M).backward()
+                total += float(loss.detach())
+            else:
+                (grads,) = self.comm.fused_send_recv(
+                    sends=[(list(out), self.next_rank)],
+                    recvs=[(self._grad_specs(out_b), self.next_rank)],
+                )
+                outs_req = [t for t in out_b if torch.is_tensor(t) and t.requires_grad]
+                torch.autograd.backward(outs_req, grads)
+
+            # send bwd grads + receive next fwd input (fused with prev rank)
+            if not self.is_first:
+                in_grads = [t.grad for t in held_b if torch.is_tensor(t) and t.requires_grad]
+                want_fwd = fwd_i < M and i + 1 < remaining
+                fwd_meta = self.comm.cached_recv_meta(self.prev_rank, "fwd") if want_fwd else None
+                res = self.comm.fused_send_recv(
+                    sends=[(in_grads, self.prev_rank)],
+                    recvs=[(fwd_meta, self.prev_rank)] if fwd_meta else [],
+                )
+                if fwd_meta:
+                    pending_input = res[0]
+                for t in held_b:
+                    if torch.is_tensor(t):
+                        t.grad = None
+            bwd_i += 1
+
+        # ---- cooldown: drain remaining backwards. NOTE: must use the
+        # fused (meta-less) transport — the steady-state fused ops never
+        # exchanged channel metadata, so the handshake-based plain path
+        # would desynchronize the control plane. ----
+        while saved:
+            held_b, out_b = saved.popleft()
+            if self.is_last:
+                logits = out_b[0] if len(out_b) == 1 else out_b
+                loss = self.loss_fn(logits, mb_labels[bwd_i].to(logits.device))
+                (loss / M).backward()
+                total += float(loss.detach())
+            else:
+                (grads,) = self.comm.fused_send_recv(
+                    sends=[], recvs=[(self._grad_specs(out_b), self.next_rank)]
+                )
+                outs_req = [t for t in out_b if torch.is_tensor(t) and t.requires_grad]
+                torch.autograd.backward(outs_req, grads)
+            if not self.is_first:
+                in_grads = [t.grad for t in held_b if torch.is_tensor(t) and t.requires_grad]
+                self.comm.fused_send_recv(sends=[(in_grads, self.prev_rank)], recvs=[])
+                for t in held_b:
+                    if torch.is_tensor(t):
+                        t.grad = None
+            bwd_i += 1
+        return total
 
     def evaluate_batch(self, inputs=None, labels=None):
         """Forward-only pass; returns logits on the last stage."""

@@ -133,3 +133,53 @@ def test_reordered_stages_and_idle_rank():
     layer_cfgs = tiny_bert_cfg(1)
     batch, labels = _make_batch()
     run_multiprocess(_uneven_plan_worker, 3, 29640, layer_cfgs, batch, labels, ".")
+
+
+def _1f1b_worker(rank, world_size, layer_cfgs, batch, labels, lr, steps, M, out_dir):
+    _pipeline_worker(rank, world_size, layer_cfgs, batch, labels, lr, steps, M, "1f1b", out_dir)
+
+
+def test_1f1b_matches_local_three_stages(tmp_path):
+    """1F1B schedule (fused crossing P2P) must produce the same losses as
+    single-process training."""
+    layer_cfgs = tiny_bert_cfg(3)
+    batch, labels = _make_batch()
+    lr, steps = 0.05, 3
+    ref = _single_process_reference(layer_cfgs, batch, labels, lr, steps, 4)
+    run_multiprocess(
+        _1f1b_3s_entry, 3, 29860, layer_cfgs, batch, labels, lr, steps, 4, str(tmp_path)
+    )
+    got = np.load(f"{tmp_path}/losses.npy")
+    assert np.allclose(got, np.array(ref), rtol=1e-4, atol=1e-5), (got, ref)
+
+
+def _1f1b_3s_entry(rank, world_size, layer_cfgs, batch, labels, lr, steps, M, out_dir):
+    torch.manual_seed(1234)
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import PartitionPlan, PipelineEngine, destroy, init_distributed
+    from skycomputing_amd.builder import build_module_from_cfg
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    full = build_module_from_cfg(layer_cfgs, record_forward_time=False)
+    L = len(layer_cfgs)
+    cuts = [0, L // 3, 2 * L // 3, L]
+    plan = PartitionPlan(stage_ranks=[0, 1, 2],
+                         ranges=[(cuts[i], cuts[i + 1]) for i in range(3)])
+    engine = PipelineEngine(comm, layer_cfgs, plan, loss_fn=torch.nn.CrossEntropyLoss(),
+                            stage_kwargs=dict(record_forward_time=False))
+    start, end = plan.ranges[engine.stage_idx]
+    engine.stage.load_layer_state_dicts(
+        [{k: v.detach().clone() for k, v in full.module[i].state_dict().items()}
+         for i in range(start, end)]
+    )
+    opt = FusedSGD(engine.parameters(), lr=lr)
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = engine.run_iteration(batch, labels, num_microbatches=M, schedule="1f1b")
+        opt.step()
+        losses.append(loss)
+    if rank == 0:
+        np.save(f"{out_dir}/losses.npy", np.array(losses, dtype=np.float64))
+    comm.barrier()
+    destroy()
