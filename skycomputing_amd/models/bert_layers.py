@@ -53,8 +53,9 @@ class SkyLayerNorm(nn.Module):
         self.bias = nn.Parameter(torch.zeros(hidden_size))
         self.eps = eps
 
-    def forward(self, x, residual=None):
-        return ops.layer_norm(x, self.weight, self.bias, self.eps, residual)
+    def forward(self, x, residual=None, dropout_p: float = 0.0, training: bool = False):
+        return ops.layer_norm(x, self.weight, self.bias, self.eps, residual,
+                              dropout_p, training)
 
 
 class LinearActivation(nn.Module):
@@ -127,8 +128,9 @@ class BertSelfOutput(nn.Module):
 
     def forward(self, hidden, residual):
         x = self.dense(hidden)
-        x = ops.dropout(x, self.dropout_p, self.training)
-        return self.layer_norm(x, residual=residual)
+        # dropout fused into the LN kernel: LN(dropout(x) + residual)
+        return self.layer_norm(x, residual=residual,
+                               dropout_p=self.dropout_p, training=self.training)
 
 
 @LAYER.register_module
@@ -251,8 +253,8 @@ class BertLayerTail(nn.Module):
 
     def forward(self, attn_out, inter, ext_mask):
         x = self.dense(inter)
-        x = ops.dropout(x, self.dropout_p, self.training)
-        hidden = self.layer_norm(x, residual=attn_out)
+        hidden = self.layer_norm(x, residual=attn_out,
+                                 dropout_p=self.dropout_p, training=self.training)
         return hidden, ext_mask
 
     def layer_flops(self, batch: int, seq: int) -> float:

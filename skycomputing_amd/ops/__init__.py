@@ -43,11 +43,20 @@ def _use_hip(t: torch.Tensor) -> bool:
     return False
 
 
-def layer_norm(x, weight, bias, eps: float = 1e-12, residual=None):
+def layer_norm(x, weight, bias, eps: float = 1e-12, residual=None,
+               dropout_p: float = 0.0, training: bool = False):
+    """LayerNorm with optional fused residual add and fused PRE-add dropout
+    on x: LN(dropout(x) + residual)."""
+    p = dropout_p if training else 0.0
     if _use_hip(x):
         from .functions import LayerNormFn
 
-        return LayerNormFn.apply(x, weight, bias, eps, residual)
+        if p > 0.0 and x.shape[-1] % 8 != 0:
+            x = dropout(x, p, True)
+            p = 0.0
+        return LayerNormFn.apply(x, weight, bias, eps, residual, p)
+    if p > 0.0:
+        x = torch.nn.functional.dropout(x, p=p, training=True)
     return eager.layer_norm(x, weight, bias, eps, residual)
 
 

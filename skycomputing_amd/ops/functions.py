@@ -50,12 +50,15 @@ def _dt(t: torch.Tensor) -> int:
 
 
 class LayerNormFn(torch.autograd.Function):
-    """Fused (residual-add +) LayerNorm, fwd+bwd in single-pass HIP kernels
-    (replaces the reference's optional apex FusedLayerNorm,
-    reference: scaelum/model/bert_layers.py:128-168)."""
+    """Fused (dropout + residual-add +) LayerNorm, fwd+bwd in single-pass
+    HIP kernels (replaces the reference's optional apex FusedLayerNorm +
+    preceding nn.Dropout, reference: scaelum/model/bert_layers.py:128-168,
+    283-288). With dropout_p > 0 the op computes
+    LN(dropout(x) + residual); the mask regenerates from (salt, device
+    step counter) in backward."""
 
     @staticmethod
-    def forward(ctx, x, weight, bias, eps, residual):
+    def forward(ctx, x, weight, bias, eps, residual, dropout_p=0.0):
         lib = hiplib.require()
         x = x.contiguous()
         residual = residual.contiguous() if residual is not None else None
@@ -64,16 +67,21 @@ class LayerNormFn(torch.autograd.Function):
         y = torch.empty_like(x)
         mean = torch.empty(rows, dtype=torch.float32, device=x.device)
         rstd = torch.empty(rows, dtype=torch.float32, device=x.device)
+        keep = 1.0 - dropout_p
+        salt = _next_seed() if keep < 1.0 else 0
+        state_ptr = rng_state().data_ptr() if keep < 1.0 else 0
         check(
             lib.sky_layernorm_fwd(
                 _stream(), ptr(x), ptr(residual), ptr(weight), ptr(bias),
                 ptr(y), ptr(mean), ptr(rstd), rows, cols, eps, _dt(x),
+                keep, salt, state_ptr,
             ),
             "sky_layernorm_fwd",
         )
         ctx.save_for_backward(x, weight, mean, rstd)
         ctx.residual = residual
         ctx.has_residual = residual is not None
+        ctx.keep, ctx.salt = keep, salt
         return y
 
     @staticmethod
@@ -94,18 +102,28 @@ class LayerNormFn(torch.autograd.Function):
             dw = torch.zeros(cols, dtype=torch.float32, device=x.device)
             db = torch.zeros(cols, dtype=torch.float32, device=x.device)
             scratch = None
+        drop = ctx.keep < 1.0
+        dres = torch.empty_like(x) if (drop and ctx.has_residual) else None
+        state_ptr = rng_state().data_ptr() if drop else 0
         check(
             lib.sky_layernorm_bwd(
                 _stream(), ptr(dy), ptr(x), ptr(ctx.residual), ptr(weight),
                 ptr(mean), ptr(rstd), ptr(dx), ptr(dw), ptr(db),
-                ptr(scratch), rows, cols, _dt(x),
+                ptr(scratch), rows, cols, _dt(x), ctx.keep, ctx.salt,
+                state_ptr, ptr(dres),
             ),
             "sky_layernorm_bwd",
         )
         if not fast:
             dw = dw.to(weight.dtype)
             db = db.to(weight.dtype)
-        return dx, dw, db, None, (dx if ctx.has_residual else None)
+        if not ctx.has_residual:
+            dgrad_res = None
+        elif drop:
+            dgrad_res = dres
+        else:
+            dgrad_res = dx
+        return dx, dw, db, None, dgrad_res, None
 
 
 class BiasGeluFn(torch.autograd.Function):

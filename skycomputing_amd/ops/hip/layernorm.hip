@@ -18,15 +18,23 @@
 
 // ---------------- forward (vectorized, wave per row) ----------------
 
-template <int DT, bool HAS_RES, int CH>
+// DROP: apply dropout to x BEFORE the residual add (the BERT pattern
+// LN(dropout(dense(x)) + residual), reference bert_layers.py:283-288);
+// the mask is regenerated from (salt,state,element index) in backward.
+template <int DT, bool HAS_RES, bool DROP, int CH>
 __global__ __launch_bounds__(256) void ln_fwd_vec_kernel(
     const void* __restrict__ x, const void* __restrict__ res,
     const void* __restrict__ w, const void* __restrict__ b,
     void* __restrict__ y, float* __restrict__ mean_out,
-    float* __restrict__ rstd_out, int64_t rows, int64_t cols, float eps) {
+    float* __restrict__ rstd_out, int64_t rows, int64_t cols, float eps,
+    float keep, uint64_t salt, const unsigned long long* __restrict__ state) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int64_t cols8 = cols / 8;
+  const uint64_t seed =
+      DROP ? salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull : 0;
+  const unsigned keep16 = keep_to_16(keep);
+  const float inv_keep = 1.f / keep;
   float v[CH][8];
   for (int64_t row = (int64_t)blockIdx.x * 4 + wid; row < rows;
        row += (int64_t)gridDim.x * 4) {
@@ -37,6 +45,17 @@ __global__ __launch_bounds__(256) void ln_fwd_vec_kernel(
       const int64_t i8 = (int64_t)c * WAVE + lane;
       if (i8 < cols8) {
         Vec8<DT>::load(x, base8 + i8, v[c]);
+        if (DROP) {
+          const uint64_t e8 = (uint64_t)(base8 + i8);
+          const uint64_t z0 = rng_hash(seed, e8 * 2);
+          const uint64_t z1 = rng_hash(seed, e8 * 2 + 1);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const uint64_t zz = j < 4 ? z0 : z1;
+            bool kb = (unsigned)((zz >> (16 * (j & 3))) & 0xFFFFu) < keep16;
+            v[c][j] = kb ? v[c][j] * inv_keep : 0.f;
+          }
+        }
         if (HAS_RES) {
           float r[8];
           Vec8<DT>::load(res, base8 + i8, r);
@@ -106,26 +125,33 @@ static inline bool ln_fast_ok(int64_t cols) {
 SKY_EXPORT int sky_layernorm_fwd(uint64_t stream, uint64_t x, uint64_t res,
                                  uint64_t w, uint64_t b, uint64_t y,
                                  uint64_t mean, uint64_t rstd, int64_t rows,
-                                 int64_t cols, float eps, int dt) {
+                                 int64_t cols, float eps, int dt, float keep,
+                                 uint64_t salt, uint64_t state) {
   hipStream_t s = (hipStream_t)stream;
   bool has_res = res != 0;
+  bool drop = keep < 1.f;
+  if (drop && !ln_fast_ok(cols)) return (int)hipErrorInvalidValue;
   if (ln_fast_ok(cols)) {
     unsigned grid = (unsigned)((rows + 3) / 4);
     if (grid > 8192u) grid = 8192u;
     const int ch = (int)((cols / 8 + WAVE - 1) / WAVE);
-#define LNFV(DT, HR, CH)                                                      \
-  hipLaunchKernelGGL((ln_fwd_vec_kernel<DT, HR, CH>), dim3(grid), dim3(256),  \
-                     0, s, (const void*)x, (const void*)res, (const void*)w,  \
-                     (const void*)b, (void*)y, (float*)mean, (float*)rstd,    \
-                     rows, cols, eps)
-#define LNFV_CH(DT, HR)                                                       \
+#define LNFV(DT, HR, DR, CH)                                                  \
+  hipLaunchKernelGGL((ln_fwd_vec_kernel<DT, HR, DR, CH>), dim3(grid),         \
+                     dim3(256), 0, s, (const void*)x, (const void*)res,       \
+                     (const void*)w, (const void*)b, (void*)y, (float*)mean,  \
+                     (float*)rstd, rows, cols, eps, keep, salt,               \
+                     (const unsigned long long*)state)
+#define LNFV_CH(DT, HR, DR)                                                   \
   do {                                                                        \
-    if (ch <= 2) LNFV(DT, HR, 2);                                             \
-    else if (ch <= 4) LNFV(DT, HR, 4);                                        \
-    else LNFV(DT, HR, 8);                                                     \
+    if (ch <= 2) LNFV(DT, HR, DR, 2);                                         \
+    else if (ch <= 4) LNFV(DT, HR, DR, 4);                                    \
+    else LNFV(DT, HR, DR, 8);                                                 \
   } while (0)
-    if (dt == DT_F32) { if (has_res) LNFV_CH(DT_F32, true); else LNFV_CH(DT_F32, false); }
-    else              { if (has_res) LNFV_CH(DT_BF16, true); else LNFV_CH(DT_BF16, false); }
+#define LNFV_D(DT, HR)                                                        \
+  do { if (drop) LNFV_CH(DT, HR, true); else LNFV_CH(DT, HR, false); } while (0)
+    if (dt == DT_F32) { if (has_res) LNFV_D(DT_F32, true); else LNFV_D(DT_F32, false); }
+    else              { if (has_res) LNFV_D(DT_BF16, true); else LNFV_D(DT_BF16, false); }
+#undef LNFV_D
 #undef LNFV_CH
 #undef LNFV
   } else {
@@ -148,16 +174,26 @@ SKY_EXPORT int sky_layernorm_fwd(uint64_t stream, uint64_t x, uint64_t res,
 // dx = rstd * (dxhat - mean_c(dxhat) - xhat * mean_c(dxhat * xhat))
 // dw[c] += sum_r dy * xhat ; db[c] += sum_r dy
 
-template <int DT, bool HAS_RES, int CH>
+// DROP: x is the PRE-dropout tensor; the mask is regenerated to rebuild
+// xs = dropout(x)+res for the statistics, dx gets the mask/keep factor,
+// and (when dres != null) the residual grad d_xs is written separately.
+template <int DT, bool HAS_RES, bool DROP, int CH>
 __global__ __launch_bounds__(256) void ln_bwd_dx_vec_kernel(
     const void* __restrict__ dy, const void* __restrict__ x,
     const void* __restrict__ res, const void* __restrict__ w,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    void* __restrict__ dx, int64_t rows, int64_t cols) {
+    void* __restrict__ dx, void* __restrict__ dres, int64_t rows,
+    int64_t cols, float keep, uint64_t salt,
+    const unsigned long long* __restrict__ state) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int64_t cols8 = cols / 8;
+  const uint64_t seed =
+      DROP ? salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull : 0;
+  const unsigned keep16 = keep_to_16(keep);
+  const float inv_keep = 1.f / keep;
   float xh[CH][8], dxh[CH][8];
+  unsigned kbits[CH];
   for (int64_t row = (int64_t)blockIdx.x * 4 + wid; row < rows;
        row += (int64_t)gridDim.x * 4) {
     const int64_t base8 = row * cols8;
@@ -169,6 +205,19 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_vec_kernel(
       if (i8 < cols8) {
         float xv[8], dyv[8], wv[8];
         Vec8<DT>::load(x, base8 + i8, xv);
+        if (DROP) {
+          const uint64_t e8 = (uint64_t)(base8 + i8);
+          const uint64_t z0 = rng_hash(seed, e8 * 2);
+          const uint64_t z1 = rng_hash(seed, e8 * 2 + 1);
+          kbits[c] = 0;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const uint64_t zz = j < 4 ? z0 : z1;
+            bool kb = (unsigned)((zz >> (16 * (j & 3))) & 0xFFFFu) < keep16;
+            kbits[c] |= (unsigned)kb << j;
+            xv[j] = kb ? xv[j] * inv_keep : 0.f;
+          }
+        }
         if (HAS_RES) {
           float r[8];
           Vec8<DT>::load(res, base8 + i8, r);
@@ -196,6 +245,12 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_vec_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           o[j] = rs * (dxh[c][j] - s1 - xh[c][j] * s2);
+        if (DROP && dres != nullptr) Vec8<DT>::store(dres, base8 + i8, o);
+        if (DROP) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            o[j] = ((kbits[c] >> j) & 1u) ? o[j] * inv_keep : 0.f;
+        }
         Vec8<DT>::store(dx, base8 + i8, o);
       }
     }
@@ -207,23 +262,39 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_vec_kernel(
 // (dw partial at [y][c], db partial at [y][cols+c]); stage 2 reduces.
 #define LN_SLABS 256
 
-template <int DT, bool HAS_RES, int BLOCK>
+template <int DT, bool HAS_RES, bool DROP, int BLOCK>
 __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_part_kernel(
     const void* __restrict__ dy, const void* __restrict__ x,
     const void* __restrict__ res, const float* __restrict__ mean,
     const float* __restrict__ rstd, float* __restrict__ scratch,
-    int64_t rows, int64_t cols8, int64_t rows_per_slab) {
+    int64_t rows, int64_t cols8, int64_t rows_per_slab, float keep,
+    uint64_t salt, const unsigned long long* __restrict__ state) {
   const int64_t c8 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
   if (c8 >= cols8) return;
   const int64_t slab = blockIdx.y;
   const int64_t r0 = slab * rows_per_slab;
   const int64_t r1 = min(rows, r0 + rows_per_slab);
+  const uint64_t seed =
+      DROP ? salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull : 0;
+  const unsigned keep16 = keep_to_16(keep);
+  const float inv_keep = 1.f / keep;
   float sw[8] = {0.f}, sb[8] = {0.f};
   for (int64_t r = r0; r < r1; ++r) {
     const float mu = mean[r], rs = rstd[r];
     float xv[8], dyv[8];
     Vec8<DT>::load(x, r * cols8 + c8, xv);
     Vec8<DT>::load(dy, r * cols8 + c8, dyv);
+    if (DROP) {
+      const uint64_t e8 = (uint64_t)(r * cols8 + c8);
+      const uint64_t z0 = rng_hash(seed, e8 * 2);
+      const uint64_t z1 = rng_hash(seed, e8 * 2 + 1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const uint64_t zz = j < 4 ? z0 : z1;
+        bool kb = (unsigned)((zz >> (16 * (j & 3))) & 0xFFFFu) < keep16;
+        xv[j] = kb ? xv[j] * inv_keep : 0.f;
+      }
+    }
     if (HAS_RES) {
       float rv[8];
       Vec8<DT>::load(res, r * cols8 + c8, rv);
@@ -333,26 +404,34 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
                                  uint64_t res, uint64_t w, uint64_t mean,
                                  uint64_t rstd, uint64_t dx, uint64_t dw,
                                  uint64_t db, uint64_t scratch, int64_t rows,
-                                 int64_t cols, int dt) {
+                                 int64_t cols, int dt, float keep,
+                                 uint64_t salt, uint64_t state,
+                                 uint64_t dres) {
   hipStream_t s = (hipStream_t)stream;
   bool has_res = res != 0;
+  bool drop = keep < 1.f;
+  if (drop && !ln_fast_ok(cols)) return (int)hipErrorInvalidValue;
   if (ln_fast_ok(cols)) {
     unsigned grid = (unsigned)((rows + 3) / 4);
     if (grid > 8192u) grid = 8192u;
     const int ch = (int)((cols / 8 + WAVE - 1) / WAVE);
-#define LNBV(DT, HR, CH)                                                       \
-  hipLaunchKernelGGL((ln_bwd_dx_vec_kernel<DT, HR, CH>), dim3(grid),           \
+#define LNBV(DT, HR, DR, CH)                                                   \
+  hipLaunchKernelGGL((ln_bwd_dx_vec_kernel<DT, HR, DR, CH>), dim3(grid),       \
                      dim3(256), 0, s, (const void*)dy, (const void*)x,         \
                      (const void*)res, (const void*)w, (const float*)mean,     \
-                     (const float*)rstd, (void*)dx, rows, cols)
-#define LNBV_CH(DT, HR)                                                        \
+                     (const float*)rstd, (void*)dx, (void*)dres, rows, cols,   \
+                     keep, salt, (const unsigned long long*)state)
+#define LNBV_CH(DT, HR, DR)                                                    \
   do {                                                                         \
-    if (ch <= 2) LNBV(DT, HR, 2);                                              \
-    else if (ch <= 4) LNBV(DT, HR, 4);                                         \
-    else LNBV(DT, HR, 8);                                                      \
+    if (ch <= 2) LNBV(DT, HR, DR, 2);                                          \
+    else if (ch <= 4) LNBV(DT, HR, DR, 4);                                     \
+    else LNBV(DT, HR, DR, 8);                                                  \
   } while (0)
-    if (dt == DT_F32) { if (has_res) LNBV_CH(DT_F32, true); else LNBV_CH(DT_F32, false); }
-    else              { if (has_res) LNBV_CH(DT_BF16, true); else LNBV_CH(DT_BF16, false); }
+#define LNBV_D(DT, HR)                                                         \
+  do { if (drop) LNBV_CH(DT, HR, true); else LNBV_CH(DT, HR, false); } while (0)
+    if (dt == DT_F32) { if (has_res) LNBV_D(DT_F32, true); else LNBV_D(DT_F32, false); }
+    else              { if (has_res) LNBV_D(DT_BF16, true); else LNBV_D(DT_BF16, false); }
+#undef LNBV_D
 #undef LNBV_CH
 #undef LNBV
   } else {
@@ -372,13 +451,17 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
     const int64_t nslabs = rows < LN_SLABS ? rows : LN_SLABS;
     const int64_t slab = (rows + nslabs - 1) / nslabs;
     dim3 grid((unsigned)((cols8 + BLOCK - 1) / BLOCK), (unsigned)nslabs);
-#define LNWBP(DT, HR)                                                          \
-  hipLaunchKernelGGL((ln_bwd_wb_part_kernel<DT, HR, BLOCK>), grid,             \
+#define LNWBP(DT, HR, DR)                                                      \
+  hipLaunchKernelGGL((ln_bwd_wb_part_kernel<DT, HR, DR, BLOCK>), grid,         \
                      dim3(BLOCK), 0, s, (const void*)dy, (const void*)x,       \
                      (const void*)res, (const float*)mean,                     \
-                     (const float*)rstd, (float*)scratch, rows, cols8, slab)
-    if (dt == DT_F32) { if (has_res) LNWBP(DT_F32, true); else LNWBP(DT_F32, false); }
-    else              { if (has_res) LNWBP(DT_BF16, true); else LNWBP(DT_BF16, false); }
+                     (const float*)rstd, (float*)scratch, rows, cols8, slab,   \
+                     keep, salt, (const unsigned long long*)state)
+#define LNWBP_D(DT, HR)                                                        \
+  do { if (drop) LNWBP(DT, HR, true); else LNWBP(DT, HR, false); } while (0)
+    if (dt == DT_F32) { if (has_res) LNWBP_D(DT_F32, true); else LNWBP_D(DT_F32, false); }
+    else              { if (has_res) LNWBP_D(DT_BF16, true); else LNWBP_D(DT_BF16, false); }
+#undef LNWBP_D
 #undef LNWBP
     dim3 g2((unsigned)((cols + 15) / 16));
     if (dt == DT_BF16)

@@ -31,10 +31,10 @@ f64 = ctypes.c_double
 def _register_signatures(lib):
     sigs = {
         # layernorm: fwd saves mean/rstd (fp32 per row) for bwd
-        "sky_layernorm_fwd": [u64, u64, u64, u64, u64, u64, u64, u64, i64, i64, f32, i32],
-        #                     strm  x   res  w    b    y    mean rstd rows cols eps  dt
-        "sky_layernorm_bwd": [u64, u64, u64, u64, u64, u64, u64, u64, u64, u64, u64, i64, i64, i32],
-        #                 strm dy x res w mean rstd dx dw db scratch | rows cols dt
+        "sky_layernorm_fwd": [u64, u64, u64, u64, u64, u64, u64, u64, i64, i64, f32, i32, f32, u64, u64],
+        #        strm x res w b y mean rstd rows cols eps dt keep salt state
+        "sky_layernorm_bwd": [u64, u64, u64, u64, u64, u64, u64, u64, u64, u64, u64, i64, i64, i32, f32, u64, u64, u64],
+        #        strm dy x res w mean rstd dx dw db scratch | rows cols dt keep salt state dres
         "sky_bias_gelu_fwd": [u64, u64, u64, u64, i64, i64, i32],
         #                     strm  x    b    y    rows cols dt
         "sky_bias_gelu_bwd": [u64, u64, u64, u64, u64, u64, u64, i64, i64, i32],

@@ -24,11 +24,24 @@ def tiny_bert_cfg(num_encoder_layers: int = 2, dropout: float = 0.0) -> list[dic
     )
 
 
-def run_multiprocess(fn, world_size: int, port: int, *args, timeout: float = 180.0):
+def run_multiprocess(fn, world_size: int, port: int, *args, timeout: float = 180.0,
+                     retries: int = 1):
     """Spawn `world_size` ranks running fn(rank, world_size, *args); raise on
-    any failure. The rendezvous port is jittered by pid to avoid TIME_WAIT
-    collisions between test runs."""
-    port = 20000 + (port + os.getpid() * 7) % 40000
+    failure. The rendezvous port is jittered by pid to avoid TIME_WAIT
+    collisions; transient rendezvous/teardown failures are retried once on
+    a fresh port (loaded CI machines occasionally drop a gloo connect)."""
+    last = None
+    for attempt in range(retries + 1):
+        p_eff = 20000 + (port + os.getpid() * 7 + attempt * 131) % 40000
+        try:
+            _run_once(fn, world_size, p_eff, args, timeout)
+            return
+        except (RuntimeError, TimeoutError) as e:
+            last = e
+    raise last
+
+
+def _run_once(fn, world_size, port, args, timeout):
     ctx = mp.get_context("spawn")
     procs = []
     for rank in range(world_size):
@@ -44,12 +57,16 @@ def run_multiprocess(fn, world_size: int, port: int, *args, timeout: float = 180
         procs.append(p)
     for p in procs:
         p.join(timeout)
-    for rank, p in enumerate(procs):
-        if p.is_alive():
-            p.terminate()
-            raise TimeoutError(f"rank {rank} timed out")
-        if p.exitcode != 0:
-            raise RuntimeError(f"rank {rank} exited with {p.exitcode}")
+    try:
+        for rank, p in enumerate(procs):
+            if p.is_alive():
+                raise TimeoutError(f"rank {rank} timed out")
+            if p.exitcode != 0:
+                raise RuntimeError(f"rank {rank} exited with {p.exitcode}")
+    finally:
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
 
 
 def _entry(fn, rank, world_size, env, args):
