@@ -262,3 +262,74 @@ def test_fused_attention_dropout_consistency():
     out0 = FusedAttentionFn.apply(qkv.detach(), None, scale, 0.0, False)
     rel = (out.float().mean() - out0.float().mean()).abs()
     assert rel < 0.05
+
+
+def test_layernorm_fused_dropout_matches_composition():
+    """The LN kernel's fused dropout uses the SAME (salt,index) hash scheme
+    as the standalone dropout kernel, so with a fixed salt
+    LN_fused(x, keep, salt) == eager_LN(dropout_kernel(x, keep, salt) + res)."""
+    torch.manual_seed(11)
+    from skycomputing_amd.ops.hiplib import check, ptr
+
+    lib = hiplib.require()
+    rows, cols = 256, 1024
+    dt = torch.bfloat16
+    x = torch.randn(rows, cols, dtype=dt, device="cuda")
+    res = torch.randn(rows, cols, dtype=dt, device="cuda")
+    w = (torch.rand(cols, device="cuda") + 0.5).to(dt)
+    b = torch.randn(cols, dtype=dt, device="cuda")
+    keep, salt = 0.7, 987654321
+    stream = torch.cuda.current_stream().cuda_stream
+
+    # reference: standalone dropout kernel with the same salt, then eager LN
+    xd = torch.empty_like(x)
+    check(lib.sky_dropout_fwd(stream, ptr(x), ptr(xd), x.numel(), keep, salt, 0, 1), "d")
+    ref = eager.layer_norm(xd.float(), w.float(), b.float(), 1e-12, res.float())
+
+    y = torch.empty_like(x)
+    mean = torch.empty(rows, dtype=torch.float32, device="cuda")
+    rstd = torch.empty_like(mean)
+    check(
+        lib.sky_layernorm_fwd(stream, ptr(x), ptr(res), ptr(w), ptr(b), ptr(y),
+                              ptr(mean), ptr(rstd), rows, cols, 1e-12, 1,
+                              keep, salt, 0),
+        "lnf",
+    )
+    torch.cuda.synchronize()
+    assert torch.allclose(y.float(), ref, atol=6e-2, rtol=6e-2), (
+        (y.float() - ref).abs().max()
+    )
+
+
+def test_layernorm_dropout_autograd_vs_reference():
+    """Full autograd path of LN(dropout(x)+res): gradients must match the
+    fp32 composition computed with the mask RECOVERED from the fused op
+    (mask = positions where fused dx is zero given res grad differs)."""
+    torch.manual_seed(12)
+    from skycomputing_amd.ops.functions import LayerNormFn
+
+    rows, cols = 128, 512
+    x = torch.randn(rows, cols, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    res = torch.randn(rows, cols, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    w = (torch.rand(cols, device="cuda") + 0.5).bfloat16().requires_grad_(True)
+    b = torch.randn(cols, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    p = 0.4
+    y = LayerNormFn.apply(x, w, b, 1e-12, res, p)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    # recover the keep mask from dx (zero exactly where dropped)
+    mask = (x.grad != 0).float()
+    keep_frac = mask.mean().item()
+    assert abs(keep_frac - (1 - p)) < 0.03
+    # fp32 composition with the recovered mask must reproduce y and grads
+    xf = x.detach().float().requires_grad_(True)
+    rf = res.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    xd = xf * mask / (1 - p)
+    yr = eager.layer_norm(xd, wf, bf, 1e-12, rf)
+    yr.backward(dy.float())
+    assert torch.allclose(y.float(), yr, atol=6e-2, rtol=6e-2)
+    assert torch.allclose(x.grad.float(), xf.grad, atol=6e-2, rtol=6e-2)
+    assert torch.allclose(res.grad.float(), rf.grad, atol=6e-2, rtol=6e-2)
+    assert torch.allclose(w.grad.float(), wf.grad, atol=0.3, rtol=0.05)
