@@ -118,15 +118,20 @@ def main():
     batches = list(loader)
 
     graphed = None
-    if use_cuda and world == 1 and sd == 0 and not args.no_graph:
-        from skycomputing_amd.parallel.graph import GraphedTrainStep
-
+    if use_cuda and sd == 0 and not args.no_graph and args.schedule == "gpipe":
         data0, labels0 = batches[0]
-        graphed = GraphedTrainStep(
-            engine.stage, opt,
-            lambda logits, labels: torch.nn.functional.cross_entropy(logits.float(), labels),
-            list(data0), labels0,
-        )
+        if world == 1 and M == 1:
+            from skycomputing_amd.parallel.graph import GraphedTrainStep
+
+            graphed = GraphedTrainStep(
+                engine.stage, opt,
+                lambda logits, labels: torch.nn.functional.cross_entropy(logits.float(), labels),
+                list(data0), labels0,
+            )
+        elif engine.stage_idx is not None:
+            from skycomputing_amd.parallel.static_exec import GraphedPipelineStep
+
+            graphed = GraphedPipelineStep(engine, opt, M, list(data0), labels0)
 
     def step(i):
         data, labels = batches[i % len(batches)]

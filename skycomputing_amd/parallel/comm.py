@@ -106,6 +106,13 @@ class CommContext:
     def cached_recv_meta(self, src: int, key: str):
         return self._meta_cache_recv.get((src, key))
 
+    def recv_tensors_into(self, bufs: list, src: int):
+        """Receive into preallocated (static) buffers — the graphed
+        pipeline executor's transport (no allocation, no meta traffic)."""
+        for b in bufs:
+            if torch.is_tensor(b):
+                dist.recv(b.detach(), src=src)
+
     def fused_send_recv(self, sends, recvs):
         """Post sends and recvs as ONE batch_isend_irecv — required when
         traffic crosses in both directions between a rank pair (1F1B steady
