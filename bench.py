@@ -131,7 +131,15 @@ def main():
         elif engine.stage_idx is not None:
             from skycomputing_amd.parallel.static_exec import GraphedPipelineStep
 
-            graphed = GraphedPipelineStep(engine, opt, M, list(data0), labels0)
+            try:
+                graphed = GraphedPipelineStep(engine, opt, M, list(data0), labels0)
+            except Exception as e:  # pragma: no cover - safety net for the
+                # unattended multi-GPU run: the eager schedule is comm-
+                # compatible with graphed peers, so a per-rank fallback is
+                # safe.
+                print(f"[bench] rank {rank}: stage-graph capture failed "
+                      f"({e!r}); falling back to eager", flush=True)
+                graphed = None
 
     def step(i):
         data, labels = batches[i % len(batches)]
