@@ -52,7 +52,6 @@ __global__ __launch_bounds__(256) void glds_probe_kernel(
     const ushort_t* __restrict__ src, ushort_t* __restrict__ dst, int mode) {
   __shared__ __attribute__((aligned(16))) char lbuf[4096];
   const int tid = threadIdx.x;
-  const int l = tid & 63;
   const int w = tid >> 6;
   const ushort_t* g = src + tid * 8;
   if (mode == 0) {

@@ -333,3 +333,59 @@ def test_layernorm_dropout_autograd_vs_reference():
     assert torch.allclose(x.grad.float(), xf.grad, atol=6e-2, rtol=6e-2)
     assert torch.allclose(res.grad.float(), rf.grad, atol=6e-2, rtol=6e-2)
     assert torch.allclose(w.grad.float(), wf.grad, atol=0.3, rtol=0.05)
+
+
+@pytest.mark.parametrize("transA,transB", [(0, 1), (0, 0), (1, 0)])
+def test_mfma_gemm_orientations(transA, transB):
+    """Hand MFMA GEMM (128x128x64 tiles): fwd NT, dgrad NN, wgrad TN."""
+    torch.manual_seed(13)
+    from skycomputing_amd.ops.hiplib import check, ptr
+
+    lib = hiplib.require()
+    M, N, K = 256, 384, 128
+    if transA == 0:
+        a = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.2
+        ar = a.float()
+    else:
+        a = torch.randn(K, M, dtype=torch.bfloat16, device="cuda") * 0.2
+        ar = a.float().t()
+    if transB == 1:
+        b = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.2
+        br = b.float().t()
+    else:
+        b = torch.randn(K, N, dtype=torch.bfloat16, device="cuda") * 0.2
+        br = b.float()
+    c = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    rc = lib.sky_gemm(
+        torch.cuda.current_stream().cuda_stream, ptr(a), ptr(b), ptr(c), 0, 0,
+        M, N, K, a.stride(0), b.stride(0), c.stride(0), transA, transB, 0, 1, 1,
+    )
+    torch.cuda.synchronize()
+    assert rc == 0
+    ref = ar @ br
+    assert torch.allclose(c.float(), ref, atol=5e-2, rtol=5e-2), (
+        (c.float() - ref).abs().max()
+    )
+
+
+def test_mfma_gemm_bias_gelu_epilogue():
+    torch.manual_seed(14)
+    from skycomputing_amd.ops.hiplib import check, ptr
+
+    lib = hiplib.require()
+    M, N, K = 128, 256, 64
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.3
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.3
+    b = torch.randn(N, dtype=torch.bfloat16, device="cuda")
+    c = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    z = torch.empty_like(c)
+    rc = lib.sky_gemm(
+        torch.cuda.current_stream().cuda_stream, ptr(x), ptr(w), ptr(c),
+        ptr(b), ptr(z), M, N, K, K, K, N, 0, 1, 2, 1, 1,
+    )
+    torch.cuda.synchronize()
+    assert rc == 0
+    zr = x.float() @ w.float().t() + b.float()
+    ref = eager.gelu(zr)
+    assert torch.allclose(z.float(), zr, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(c.float(), ref, atol=5e-2, rtol=5e-2)
