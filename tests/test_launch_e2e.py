@@ -175,3 +175,83 @@ def test_resnet_pipeline_builds_and_runs():
     out = stage(x)
     assert out.shape == (2, 10)
     out.sum().backward()
+
+
+def _replan_worker(rank, world, layer_cfgs, batch, labels, out_dir):
+    """Train 2 iters on plan A, migrate weights through the ParameterServer
+    to a different partition, continue training — loss sequence must be
+    identical to training on a fixed plan (same math, different placement)."""
+    torch.manual_seed(77)
+    from skycomputing_amd.dynamics import ParameterServer
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import PartitionPlan, PipelineEngine, destroy, init_distributed
+    from skycomputing_amd.builder import build_module_from_cfg
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    L = len(layer_cfgs)
+    full = build_module_from_cfg(layer_cfgs, record_forward_time=False)
+
+    def fresh_engine(plan):
+        e = PipelineEngine(comm, layer_cfgs, plan,
+                           loss_fn=torch.nn.CrossEntropyLoss(),
+                           stage_kwargs=dict(record_forward_time=False))
+        return e
+
+    def seed_from_full(e):
+        s0, e0 = e.plan.ranges[e.stage_idx]
+        e.stage.load_layer_state_dicts(
+            [{k: v.detach().clone() for k, v in full.module[i].state_dict().items()}
+             for i in range(s0, e0)]
+        )
+
+    lr = 0.05
+    planA = PartitionPlan(stage_ranks=[0, 1], ranges=[(0, L // 2), (L // 2, L)])
+    planB = PartitionPlan(stage_ranks=[1, 0], ranges=[(0, 2), (2, L)])
+
+    # run: 2 iters on A -> migrate -> 2 iters on B
+    eng = fresh_engine(planA)
+    seed_from_full(eng)
+    opt = FusedSGD(eng.parameters(), lr=lr)
+    losses = []
+    for _ in range(2):
+        opt.zero_grad()
+        losses.append(eng.run_iteration(batch, labels, 2, "gpipe"))
+        opt.step()
+    ps = ParameterServer(L)
+    ps.gather_from_engine(eng, comm)
+    eng2 = fresh_engine(planB)
+    ps.scatter_to_engine(eng2, comm)
+    opt2 = FusedSGD(eng2.parameters(), lr=lr)
+    for _ in range(2):
+        opt2.zero_grad()
+        losses.append(eng2.run_iteration(batch, labels, 2, "gpipe"))
+        opt2.step()
+
+    # reference: 4 iters on the fixed plan A
+    eng3 = fresh_engine(planA)
+    seed_from_full(eng3)
+    opt3 = FusedSGD(eng3.parameters(), lr=lr)
+    ref = []
+    for _ in range(4):
+        opt3.zero_grad()
+        ref.append(eng3.run_iteration(batch, labels, 2, "gpipe"))
+        opt3.step()
+
+    if rank == 0:
+        import numpy as np
+
+        assert np.allclose(losses, ref, rtol=1e-4, atol=1e-5), (losses, ref)
+    comm.barrier()
+    destroy()
+
+
+def test_reallocation_mid_training(tmp_path):
+    from .helpers import run_multiprocess, tiny_bert_cfg
+    import torch as _t
+
+    layer_cfgs = tiny_bert_cfg(2)
+    g = _t.Generator().manual_seed(0)
+    ids = _t.randint(0, 500, (8, 16), generator=g)
+    batch = (ids, _t.zeros_like(ids), _t.ones_like(ids))
+    labels = _t.randint(0, 3, (8,), generator=g)
+    run_multiprocess(_replan_worker, 2, 29910, layer_cfgs, batch, labels, str(tmp_path))
