@@ -27,7 +27,7 @@ data_config = dict(
     dataset=dict(layer_type="SyntheticGlueDataset", size=64, max_seq_length=16,
                  vocab_size=500, num_class=3, seed=1),
 )
-worker_config = [dict(slowdown=0.0), dict(slowdown=1.0)]
+worker_config = [dict(slowdown=0.0), dict(slowdown=4.0)]
 allocator_config = dict(
     mode="optimal",
     benchmark=dict(batch_size=4, seq_len=16, hidden=64, iterations=2),
@@ -45,7 +45,7 @@ logging_config = dict(log_dir="{logdir}")
 
 def test_launch_cli_two_ranks_optimal(tmp_path):
     """Full driver path under torch.distributed.run: benchmark -> optimal
-    allocation (rank 1 slowed 1x) -> pipeline training with hooks."""
+    allocation (rank 1 slowed 4x) -> pipeline training with hooks."""
     cfg = tmp_path / "cfg.py"
     cfg.write_text(TINY_CONFIG.format(logdir=str(tmp_path / "logs")))
     env = dict(os.environ)
