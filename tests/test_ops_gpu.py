@@ -389,3 +389,41 @@ def test_mfma_gemm_bias_gelu_epilogue():
     ref = eager.gelu(zr)
     assert torch.allclose(z.float(), zr, atol=5e-2, rtol=5e-2)
     assert torch.allclose(c.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_attention_fused_vs_decomposed_paths(monkeypatch):
+    """ops.attention: the fused kernel and the decomposed
+    (bmm + masked-softmax) fallback must agree (no dropout)."""
+    torch.manual_seed(15)
+    from skycomputing_amd import ops
+
+    qkv = torch.randn(2, 128, 3, 4, 64, dtype=torch.bfloat16, device="cuda")
+    mask = torch.zeros(2, 1, 1, 128, dtype=torch.bfloat16, device="cuda")
+    mask[:, :, :, 100:] = -10000.0
+    fused = ops.attention(qkv, mask, 0.0, False)
+    monkeypatch.setenv("SKY_NO_FUSED_ATTN", "1")
+    decomposed = ops.attention(qkv, mask, 0.0, False)
+    assert torch.allclose(fused.float(), decomposed.float(), atol=6e-2, rtol=6e-2), (
+        (fused.float() - decomposed.float()).abs().max()
+    )
+
+
+def test_attention_decomposed_backward(monkeypatch):
+    """The env-gated decomposed backward (probs recompute + bmms) stays
+    correct — it is the fallback if the fused backward is disabled."""
+    torch.manual_seed(16)
+    from skycomputing_amd.ops.functions import FusedAttentionFn
+
+    monkeypatch.setenv("SKY_NO_FUSED_ATTN_BWD", "1")
+    qkv = torch.randn(2, 64, 3, 4, 64, dtype=torch.bfloat16, device="cuda",
+                      requires_grad=True)
+    out = FusedAttentionFn.apply(qkv, None, 0.125, 0.0, False)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    qf = qkv.detach().float().requires_grad_(True)
+    q = qf[:, :, 0].permute(0, 2, 1, 3)
+    k = qf[:, :, 1].permute(0, 2, 1, 3)
+    v = qf[:, :, 2].permute(0, 2, 1, 3)
+    ref = eager.attention_context(q, k, v, None).permute(0, 2, 1, 3)
+    ref.backward(dout.float())
+    assert torch.allclose(qkv.grad.float(), qf.grad, atol=8e-2, rtol=8e-2)
