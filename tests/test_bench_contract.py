@@ -1,0 +1,40 @@
+"""Driver-contract guard: bench.py must emit exactly one JSON line with the
+required schema from a plain single-process invocation."""
+
+from __future__ import annotations
+
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_json_contract(tmp_path):
+    out = tmp_path / "b.json"
+    env = dict(os.environ)
+    env["MASTER_PORT"] = str(21000 + os.getpid() % 20000)
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--layers", "1", "--batch", "8",
+         "--seq", "16", "--steps", "2", "--warmup", "1",
+         "--json-out", str(out)],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=420,
+    )
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    json_lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1
+    d = json.loads(json_lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in d, key
+    assert d["higher_is_better"] is False
+    assert d["scaling"] == "strong"
+    assert d["data"] == "synthetic"
+    assert d["n_gpus"] == 1 and d["steps"] == 2 and d["warmup"] == 1
+    assert d["value"] > 0 and abs(d["ms_per_step"] - d["value"] * 1e3) < 1e-6
+    cfg = d["config"]
+    for key in ("model", "global_batch", "seq_len", "parallelism"):
+        assert key in cfg, key
+    assert json.loads(out.read_text()) == d

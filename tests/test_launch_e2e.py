@@ -135,10 +135,6 @@ def _stop_worker(rank, world, layer_cfgs, root):
     runner = Runner(engine, opt, comm, max_epoch=1, max_iter=50, log_interval=100)
     runner.register_hook(StopHook(root=root))
 
-    class FlagAtIter2:
-        def __init__(self):
-            self.fired = False
-
     def flip(r):
         if r.iter == 1 and rank == 0:
             StopHook.stop(root)
