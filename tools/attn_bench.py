@@ -62,13 +62,12 @@ print(f"torch sdpa (no drop): "
 dout = torch.randn(B, S, h, d, dtype=torch.bfloat16, device="cuda")
 pdT = torch.empty(B, h, S, S, dtype=torch.bfloat16, device="cuda")
 dsT = torch.empty_like(pdT)
-dsR = torch.empty_like(pdT)
 dqkv = torch.empty_like(qkv)
 
 
 def run_bwd(keep=0.9):
     check(lib.sky_attn_bwd(stream, ptr(qkv), ptr(dout), ptr(mask), ptr(m),
-                           ptr(l), ptr(pdT), ptr(dsT), ptr(dsR), ptr(dqkv),
+                           ptr(l), ptr(pdT), ptr(dsT), ptr(dqkv),
                            B, S, h, d, scale, keep, 123, 0), "bwd")
 
 
