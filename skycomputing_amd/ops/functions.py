@@ -310,12 +310,11 @@ class FusedAttentionFn(torch.autograd.Function):
             alloc = torch.empty if S == 128 else torch.zeros
             pdT = alloc((B, h, S, S), dtype=qkv.dtype, device=dev)
             dsT = alloc((B, h, S, S), dtype=qkv.dtype, device=dev)
-            dsR = alloc((B, h, S, S), dtype=qkv.dtype, device=dev)
             dqkv = torch.empty_like(qkv)
             check(
                 lib.sky_attn_bwd(
                     _stream(), ptr(qkv), ptr(dout), ptr(mask), ptr(m),
-                    ptr(lsum), ptr(pdT), ptr(dsT), ptr(dsR), ptr(dqkv),
+                    ptr(lsum), ptr(pdT), ptr(dsT), ptr(dqkv),
                     B, S, h, d, ctx.scale, ctx.keep, ctx.salt,
                     rng_state().data_ptr(),
                 ),

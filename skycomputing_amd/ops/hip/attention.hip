@@ -411,18 +411,15 @@ SKY_EXPORT int sky_attn_probs(uint64_t stream, uint64_t qkv, uint64_t mask,
 // ============================================================================
 
 template <bool HAS_MASK>
-__global__ __launch_bounds__(256, 3) void attn_bwd1_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd1_kernel(
     const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ dout,
     const ushort_t* __restrict__ mask, const float* __restrict__ m_io,
     const float* __restrict__ l_io, ushort_t* __restrict__ pdT,
-    ushort_t* __restrict__ dsT, ushort_t* __restrict__ dsR,
-    ushort_t* __restrict__ dqkv, int B, int S,
+    ushort_t* __restrict__ dsT, ushort_t* __restrict__ dqkv, int B, int S,
     int h, float scale, float keep, uint64_t salt,
     const unsigned long long* __restrict__ state) {
   // LDS: V [128][64] sw7 @0 (16K), K [128][64] sw7 @16K, Kt [64][128] sw15
-  // @32K. Total 48K -> 3 workgroups/CU. dS also goes to a row-major
-  // GLOBAL scratch (dsR); the dQ phase reads back only this wave's own
-  // rows (same-wave RAW, drained with vmcnt(0)).
+  // @32K, per-wave dSrow [32][128] sw15 @48K+w*8K. Total 80K.
   extern __shared__ __attribute__((aligned(16))) char lds[];
   const int tid = threadIdx.x;
   const int l = tid & 63;
@@ -445,7 +442,7 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1_kernel(
   const int dots = h * ATT_D;  // dout token stride
 
   if (S < ATT_SMAX) {
-    for (int u = tid; u < (48 * 1024) / 16; u += 256)
+    for (int u = tid; u < (80 * 1024) / 16; u += 256)
       *(ushort8_t*)lds_at(lds, u * 16) = (ushort8_t)(ushort_t)0;
     __syncthreads();
   }
@@ -572,8 +569,9 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1_kernel(
           dp = kbit ? dp * inv_keep : 0.f;
         }
         const float ds = scale * p * (dp - dot[r]);
-        if (row < S)
-          dsR[((size_t)bh * S + row) * S + col] = f32_to_bf16(col < S ? ds : 0.f);
+        const int rl = qi * 16 + lg * 4 + r;
+        *(ushort_t*)lds_at(lds, swz(49152 + w * 8192 + rl * 256 + col * 2, rl, 15)) =
+            f32_to_bf16(ds);
         ds4[r] = f32_to_bf16(ds);
         pd4[r] = f32_to_bf16(pd);
       }
@@ -583,19 +581,16 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1_kernel(
         *(ushort4_t*)(pdT + o) = pd4;
       }
     }
-    // drain the dsR stores before this wave reads its own rows back
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    // dQ = dS K : A = this wave's own dS rows read back from dsR, B = Kt
+    // dQ = dS K : A = dSrow (this wave's LDS), B = Kt
     f32x4 qacc[4];
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct) qacc[ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
       if (ks * 32 >= S) break;
-      int arow = qtok_base + lm;
-      if (arow >= S) arow = S - 1;
-      bf16x8 asr = *(const bf16x8*)(dsR + ((size_t)bh * S + arow) * S +
-                                    ks * 32 + lg * 8);
+      const int rl = qi * 16 + lm;
+      bf16x8 asr = *(const bf16x8*)lds_at(
+          lds, swz(49152 + w * 8192 + rl * 256 + (ks * 32 + lg * 8) * 2, rl, 15));
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
         const int c = ct * 16 + lm;
@@ -705,8 +700,7 @@ __global__ __launch_bounds__(256) void attn_bwd2_kernel(
 
 SKY_EXPORT int sky_attn_bwd(uint64_t stream, uint64_t qkv, uint64_t dout,
                             uint64_t mask, uint64_t m, uint64_t lsum,
-                            uint64_t pdT, uint64_t dsT, uint64_t dsR,
-                            uint64_t dqkv,
+                            uint64_t pdT, uint64_t dsT, uint64_t dqkv,
                             int64_t B, int64_t S, int64_t h, int64_t d,
                             float scale, float keep, uint64_t salt,
                             uint64_t state) {
@@ -715,19 +709,17 @@ SKY_EXPORT int sky_attn_bwd(uint64_t stream, uint64_t qkv, uint64_t dout,
   dim3 grid((unsigned)(B * h));
   bool hm = mask != 0;
   if (hm)
-    hipLaunchKernelGGL((attn_bwd1_kernel<true>), grid, dim3(256), 48 * 1024, s,
+    hipLaunchKernelGGL((attn_bwd1_kernel<true>), grid, dim3(256), 80 * 1024, s,
                        (const ushort_t*)qkv, (const ushort_t*)dout,
                        (const ushort_t*)mask, (const float*)m,
                        (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
-                       (ushort_t*)dsR,
                        (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
                        salt, (const unsigned long long*)state);
   else
-    hipLaunchKernelGGL((attn_bwd1_kernel<false>), grid, dim3(256), 48 * 1024, s,
+    hipLaunchKernelGGL((attn_bwd1_kernel<false>), grid, dim3(256), 80 * 1024, s,
                        (const ushort_t*)qkv, (const ushort_t*)dout,
                        (const ushort_t*)mask, (const float*)m,
                        (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
-                       (ushort_t*)dsR,
                        (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
                        salt, (const unsigned long long*)state);
   hipLaunchKernelGGL(attn_bwd2_kernel, grid, dim3(256), 32 * 1024, s,
