@@ -54,7 +54,6 @@ def build_module_from_cfg(
     slowdown: float = 0.0,
     mem_limit: int | None = None,
     record_forward_time: bool = True,
-    checkpoint_activations: bool = False,
     **stage_kwargs,
 ) -> StageModule:
     """Layer-config list -> SequentialWrapper -> StageModule
@@ -68,7 +67,6 @@ def build_module_from_cfg(
         slowdown=slowdown,
         mem_limit=mem_limit,
         record_forward_time=record_forward_time,
-        checkpoint_activations=checkpoint_activations,
         **stage_kwargs,
     )
 

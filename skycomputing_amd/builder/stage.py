@@ -81,7 +81,6 @@ class StageModule(nn.Module):
         slowdown: float = 0.0,
         mem_limit: int | None = None,
         record_forward_time: bool = True,
-        checkpoint_activations: bool = False,
     ):
         super().__init__()
         if not isinstance(module, SequentialWrapper):
@@ -90,7 +89,6 @@ class StageModule(nn.Module):
         self.slowdown = float(slowdown)
         self.mem_limit = mem_limit
         self.record_forward_time = record_forward_time
-        self.checkpoint_activations = checkpoint_activations
         self.forward_time: list[float] = []
         self.last_forward_time = 0.0
         self._timer = DeviceTimer(use_cuda=torch.cuda.is_available())

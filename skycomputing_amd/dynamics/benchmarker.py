@@ -24,7 +24,6 @@ import os
 import torch
 
 from ..builder import build_layer, build_module_from_cfg
-from ..dataset.data_generator import RandomTokenGenerator
 from ..stimulator import Stimulator
 from .estimator import Estimator
 from .worker import WorkerManager
