@@ -183,3 +183,31 @@ def _1f1b_3s_entry(rank, world_size, layer_cfgs, batch, labels, lr, steps, M, ou
         np.save(f"{out_dir}/losses.npy", np.array(losses, dtype=np.float64))
     comm.barrier()
     destroy()
+
+
+def _val_worker(rank, world_size, layer_cfgs, out_dir):
+    torch.manual_seed(21)
+    from skycomputing_amd.dataset import SyntheticGlueDataset
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import PartitionPlan, PipelineEngine, destroy, init_distributed
+    from skycomputing_amd.runner import Runner
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    L = len(layer_cfgs)
+    plan = PartitionPlan(stage_ranks=[0, 1], ranges=[(0, L // 2), (L // 2, L)])
+    engine = PipelineEngine(comm, layer_cfgs, plan,
+                            loss_fn=torch.nn.CrossEntropyLoss(),
+                            stage_kwargs=dict(record_forward_time=False))
+    opt = FusedSGD(engine.parameters(), lr=0.01)
+    runner = Runner(engine, opt, comm, max_epoch=1, max_iter=2, log_interval=10)
+    ds = SyntheticGlueDataset(size=32, max_seq_length=16, vocab_size=500, seed=9)
+    loader = torch.utils.data.DataLoader(ds, batch_size=8, drop_last=True)
+    runner.train(loader)
+    acc = runner.val(loader, max_batches=2)
+    assert acc is not None and 0.0 <= acc <= 1.0
+    comm.barrier()
+    destroy()
+
+
+def test_runner_val_pipeline():
+    run_multiprocess(_val_worker, 2, 29950, tiny_bert_cfg(1), ".")
