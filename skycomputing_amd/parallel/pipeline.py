@@ -143,17 +143,22 @@ class PipelineEngine:
 
     # ---------------- micro-step primitives ----------------
 
-    def _forward_microbatch(self, inputs):
-        """Run one microbatch through this stage; returns (held_inputs, outputs)."""
+    def _forward_microbatch(self, inputs, key: str = "fwd"):
+        """Run one microbatch through this stage; returns (held_inputs, outputs).
+
+        ``key`` separates channel metadata per context: eval-mode outputs
+        carry different requires_grad flags than training, so they must not
+        share the cached "fwd" channel (a shared key desynchronizes the
+        control-plane handshake)."""
         if self.is_first:
             args = _to_tuple(inputs)
             held = ()
         else:
-            args = tuple(self.comm.recv_tensors(self.prev_rank, "fwd"))
+            args = tuple(self.comm.recv_tensors(self.prev_rank, key))
             held = args
         out = _to_tuple(self.stage(*args))
         if not self.is_last:
-            self.comm.send_tensors(list(out), self.next_rank, "fwd")
+            self.comm.send_tensors(list(out), self.next_rank, key)
         return held, out
 
     def _backward_microbatch(self, held, out, labels=None, num_microbatches=1):
@@ -321,7 +326,7 @@ class PipelineEngine:
         if self.stage_idx is None:
             return None
         with torch.no_grad():
-            _, out = self._forward_microbatch(inputs)
+            _, out = self._forward_microbatch(inputs, key="fwd_eval")
         return out[0] if self.is_last else None
 
     def _broadcast_loss(self, loss_val):
