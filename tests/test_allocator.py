@@ -100,3 +100,57 @@ def test_speedup_vs_even_on_synthetic_heterogeneity():
     ce = a.plan_cost(a.even_allocate())
     co = a.plan_cost(a.optimal_allocate())
     assert ce / co > 1.5  # >50% faster
+
+
+# ---------------- property-based checks (hypothesis) ----------------
+try:
+    from hypothesis import given, settings, strategies as st
+
+    HAVE_HYP = True
+except ImportError:  # pragma: no cover
+    HAVE_HYP = False
+
+
+if HAVE_HYP:
+
+    @settings(max_examples=60, deadline=None)
+    @given(
+        L=st.integers(4, 60),
+        W=st.integers(2, 8),
+        seed=st.integers(0, 10_000),
+        hetero=st.floats(1.0, 8.0),
+    )
+    def test_optimal_dominates_randomized(L, W, seed, hetero):
+        rng = random.Random(seed)
+        flops = [rng.uniform(0.1, 2.0) for _ in range(L)]
+        mem = [rng.uniform(0.5, 1.5) for _ in range(L)]
+        total_mem = sum(mem)
+        workers = [
+            dict(rank=r, time=rng.uniform(1.0, hetero),
+                 avai_mem=rng.uniform(total_mem / W * 1.2, total_mem))
+            for r in range(W)
+        ]
+        a = Allocator(flops, mem, workers)
+        try:
+            pe = a.even_allocate()
+            ce = a.plan_cost(pe)
+        except AllocationError:
+            ce = None
+        try:
+            pd = a.dynamic_allocate()
+            _check_plan(pd, L)
+            cd = a.plan_cost(pd)
+        except AllocationError:
+            cd = None
+        po = a.optimal_allocate()
+        _check_plan(po, L)
+        co = a.plan_cost(po)
+        # memory feasibility of the optimal plan
+        by_rank = {w["rank"]: w for w in a.workers}
+        for r, (s, e) in zip(po.stage_ranks, po.ranges):
+            assert sum(mem[s:e]) <= by_rank[r]["avai_mem"] * (1 + 1e-9)
+        # optimal dominates the heuristics whenever they are feasible
+        if cd is not None:
+            assert co <= cd * (1 + 1e-6)
+        if ce is not None:
+            assert co <= ce * (1 + 1e-6)
