@@ -117,12 +117,6 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   }
   __syncthreads();
 
-  f32x4 oacc[2][4];
-#pragma unroll
-  for (int qi = 0; qi < 2; ++qi)
-#pragma unroll
-    for (int dv = 0; dv < 4; ++dv) oacc[qi][dv] = (f32x4){0.f, 0.f, 0.f, 0.f};
-
 #pragma clang loop unroll(disable)
   for (int qi = 0; qi < 2; ++qi) {
     const int qtile = qt0 + qi;
@@ -247,6 +241,14 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   }
 
   if (PROBS_MODE) return;
+
+  // PV accumulators live only from here (declaring them earlier keeps 32
+  // VGPRs hot through the scores/softmax phase and forces spills)
+  f32x4 oacc[2][4];
+#pragma unroll
+  for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+    for (int dv = 0; dv < 4; ++dv) oacc[qi][dv] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
   // ---- stage V transposed over K's (now dead) region ----
   __syncthreads();  // all waves done reading K and writing their P tiles
