@@ -42,9 +42,14 @@ class Allocator:
     ):
         """``workers``: [{'rank': r, 'time': dt, 'avai_mem': bytes}, ...] —
         dt is the probe time (relative speed; higher = slower).
-        ``boundary_cost``: optional per-cut payload estimate (len L-1) for
-        comm-aware balancing (penalizes e.g. Body->Tail cuts whose hop
-        carries the 4x intermediate tensor)."""
+        ``boundary_cost``: optional per-cut payload estimate (len L-1),
+        weighted by ``comm_weight``, for comm-aware balancing (penalizes
+        e.g. Body->Tail cuts whose hop carries the 4x intermediate tensor,
+        SURVEY.md §2c C4). It enters ``_chunk_time`` and therefore the
+        dynamic allocator's objective and ``plan_cost``; the exact optimal
+        solver currently optimizes pure compute time (its binary-search
+        extension step needs monotone chunk cost) — run dynamic_allocate
+        on top of an optimal plan when cut costs matter."""
         self.L = len(layer_flops)
         assert len(layer_mem) == self.L
         self.flops = [float(f) for f in layer_flops]

@@ -154,3 +154,36 @@ if HAVE_HYP:
             assert co <= cd * (1 + 1e-6)
         if ce is not None:
             assert co <= ce * (1 + 1e-6)
+
+
+def test_comm_aware_dynamic_avoids_expensive_cut():
+    """With a huge payload on one boundary, the comm-aware dynamic
+    allocator shifts the cut away from it."""
+    from skycomputing_amd.dynamics import bert_boundary_payloads
+
+    L, W = 12, 2
+    flops = [1.0] * L
+    mem = [0.0] * L
+    workers = [dict(rank=r, time=1.0, avai_mem=100.0) for r in range(W)]
+    cost = [0.0] * (L - 1)
+    cost[5] = 1000.0  # even split would cut exactly here
+    a0 = Allocator(flops, mem, workers)
+    a1 = Allocator(flops, mem, workers, boundary_cost=cost, comm_weight=1.0)
+    p0 = a0.dynamic_allocate()
+    p1 = a1.dynamic_allocate()
+    assert p0.ranges[0][1] == 6          # unaware: cuts at the expensive edge
+    assert p1.ranges[0][1] != 6, p1      # aware: moves the cut
+    assert a1.plan_cost(p1) < a1.plan_cost(p0)
+
+
+def test_bert_boundary_payloads_shape():
+    from skycomputing_amd.dynamics import bert_boundary_payloads
+    from skycomputing_amd.models import bert_pipeline_config
+
+    cfgs = bert_pipeline_config(2, dict(hidden_size=64, intermediate_size=256))
+    pay = bert_boundary_payloads(cfgs, batch=4, seq=8)
+    assert len(pay) == len(cfgs) - 1
+    # cuts after a Body carry hidden+intermediate (the 4x hop)
+    body_idx = [i for i, c in enumerate(cfgs) if c["layer_type"] == "BertLayer_Body"]
+    for i in body_idx:
+        assert pay[i] > pay[i - 1]
