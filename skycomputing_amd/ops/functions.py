@@ -266,9 +266,12 @@ class DropoutFn(torch.autograd.Function):
 class FusedAttentionFn(torch.autograd.Function):
     """Fused attention: out = dropout(softmax(scale*QK^T + mask)) V in ONE
     MFMA kernel (ops/hip/attention.hip); S x S probabilities are never
-    materialized in forward (row max/sum saved, flash-style). Backward
-    recomputes P (+ dropout mask) with sky_attn_probs, then runs
-    dV/dP/dS/dQ/dK as hipBLASLt batched GEMMs + the HIP softmax-backward.
+    materialized in forward (row max/sum saved, flash-style).
+
+    Backward default: TWO fused MFMA kernels (sky_attn_bwd) — recompute +
+    dS + dQ, then dK/dV over transposed scratch — no torch ops at all.
+    SKY_NO_FUSED_ATTN_BWD=1 selects the decomposed fallback (sky_attn_probs
+    recompute + hipBLASLt batched GEMMs + HIP softmax-backward).
 
     qkv: [B, S, 3, h, d] (d = 64), mask: [B, 1, 1, S] additive or None.
     Returns [B, S, h, d].
