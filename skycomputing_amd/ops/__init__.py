@@ -120,13 +120,18 @@ def attention(qkv, mask, dropout_p: float = 0.0, training: bool = False):
     B, S, three, h, d = qkv.shape
     scale = 1.0 / math.sqrt(d)
     if (
-        qkv.is_cuda and qkv.dtype == torch.bfloat16 and d == 64 and S <= 128
+        qkv.is_cuda and qkv.dtype == torch.bfloat16 and d == 64
         and hiplib.available() and os.environ.get("SKY_NO_FUSED_ATTN") != "1"
     ):
-        from .functions import FusedAttentionFn
+        from .functions import FlashAttentionFn, FusedAttentionFn
 
-        out = FusedAttentionFn.apply(qkv, mask, scale, dropout_p, training)
-        return out.reshape(B, S, h * d)
+        if S <= 128:
+            out = FusedAttentionFn.apply(qkv, mask, scale, dropout_p, training)
+            return out.reshape(B, S, h * d)
+        if S <= 4096:
+            # flash-style forward (online softmax); decomposed backward
+            out = FlashAttentionFn.apply(qkv, mask, scale, dropout_p, training)
+            return out.reshape(B, S, h * d)
     q = qkv[:, :, 0].permute(0, 2, 1, 3)
     k = qkv[:, :, 1].permute(0, 2, 1, 3)
     v = qkv[:, :, 2].permute(0, 2, 1, 3)

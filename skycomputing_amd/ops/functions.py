@@ -369,6 +369,101 @@ class FusedAttentionFn(torch.autograd.Function):
         return dqkv, None, None, None, None
 
 
+class FlashAttentionFn(torch.autograd.Function):
+    """Flash-style attention for arbitrary S (d = 64, bf16): the forward is
+    one online-softmax MFMA kernel per (batch, head, 128-query block); the
+    S x S matrix never exists in forward. Backward recomputes probabilities
+    (hipBLASLt scores GEMM + HIP masked-softmax) and regenerates the
+    dropout mask from the saved salt (linear element indices match the
+    generic dropout kernels), then runs the standard five-GEMM backward."""
+
+    @staticmethod
+    def forward(ctx, qkv, mask, scale, dropout_p, training):
+        lib = hiplib.require()
+        B, S, three, h, d = qkv.shape
+        assert three == 3 and d == 64
+        qkv = qkv.contiguous()
+        mask = mask.contiguous() if mask is not None else None
+        out = torch.empty(B, S, h, d, dtype=qkv.dtype, device=qkv.device)
+        m = torch.empty(B, h, S, dtype=torch.float32, device=qkv.device)
+        lsum = torch.empty_like(m)
+        keep = 1.0 - dropout_p if (dropout_p > 0 and training) else 1.0
+        salt = _next_seed() if keep < 1.0 else 0
+        check(
+            lib.sky_attn_flash_fwd(
+                _stream(), ptr(qkv), ptr(mask), ptr(out), ptr(m), ptr(lsum),
+                B, S, h, d, scale, keep, salt, rng_state().data_ptr(),
+            ),
+            "sky_attn_flash_fwd",
+        )
+        ctx.save_for_backward(qkv, mask)
+        ctx.scale, ctx.keep, ctx.salt = scale, keep, salt
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        lib = hiplib.require()
+        qkv, mask = ctx.saved_tensors
+        B, S, _, h, d = qkv.shape
+        dev = qkv.device
+        q = qkv[:, :, 0].permute(0, 2, 1, 3)
+        k = qkv[:, :, 1].permute(0, 2, 1, 3)
+        v = qkv[:, :, 2].permute(0, 2, 1, 3)
+        scores = torch.matmul(q, k.transpose(-1, -2)).contiguous()
+        P = torch.empty_like(scores)
+        check(
+            lib.sky_masked_softmax_fwd(
+                _stream(), ptr(scores), ptr(mask), ptr(P), B, h, S, S,
+                ctx.scale, 1.0, 0, _dt(scores),
+            ),
+            "sky_masked_softmax_fwd",
+        )
+        if ctx.keep < 1.0:
+            Pd = torch.empty_like(P)
+            check(
+                lib.sky_dropout_fwd(
+                    _stream(), ptr(P), ptr(Pd), P.numel(), ctx.keep, ctx.salt,
+                    rng_state().data_ptr(), _dt(P),
+                ),
+                "sky_dropout_fwd",
+            )
+        else:
+            Pd = P
+        dO = dout.permute(0, 2, 1, 3).contiguous()
+        dV = torch.matmul(Pd.transpose(-1, -2), dO)
+        dPd = torch.matmul(dO, v.transpose(-1, -2)).contiguous()
+        if ctx.keep < 1.0:
+            dP = torch.empty_like(dPd)
+            check(
+                lib.sky_dropout_bwd(
+                    _stream(), ptr(dPd), ptr(dP), dPd.numel(), ctx.keep,
+                    ctx.salt, rng_state().data_ptr(), _dt(dPd),
+                ),
+                "sky_dropout_bwd",
+            )
+        else:
+            dP = dPd
+        dS = torch.empty_like(dP)
+        check(
+            lib.sky_masked_softmax_bwd(
+                _stream(), ptr(dP), ptr(P), ptr(dS), B, h, S, S, ctx.scale,
+                1.0, 0, _dt(P),
+            ),
+            "sky_masked_softmax_bwd",
+        )
+        dQ = torch.matmul(dS, k)
+        dK = torch.matmul(dS.transpose(-1, -2), q)
+        dqkv = torch.empty_like(qkv)
+        check(
+            lib.sky_pack3(
+                _stream(), ptr(dQ.contiguous()), ptr(dK.contiguous()),
+                ptr(dV.contiguous()), ptr(dqkv), B, S, h, d, _dt(qkv),
+            ),
+            "sky_pack3",
+        )
+        return dqkv, None, None, None, None
+
+
 def colsum(src: torch.Tensor, out_dtype=None) -> torch.Tensor:
     """Column sum over a 2D [rows, cols] tensor via the HIP column-parallel
     reduction (fp32 accumulation)."""

@@ -731,3 +731,263 @@ SKY_EXPORT int sky_attn_bwd(uint64_t stream, uint64_t qkv, uint64_t dout,
   LAUNCH_CHECK();
   return 0;
 }
+
+// ============================================================================
+// Flash-style attention forward for ARBITRARY sequence length (d = 64).
+//
+// Online-softmax over 128-key kv tiles: one workgroup per
+// (batch, head, 128-query block); running row max m / row sum l with a
+// rescale of the O accumulator at every tile (plain rescale, no defer —
+// the guide's T13 hazard does not apply). The S x S matrix never exists;
+// row stats are saved so the Python backward can recompute probabilities.
+// Dropout uses LINEAR element indices ((bh*S+row)*S+col) so the generic
+// dropout kernels regenerate the same mask in the decomposed backward.
+//
+// LDS reuses the S<=128 kernel's 48 KB layout per kv tile: K staged via
+// global_load_lds with a source-side XOR swizzle, V^T aliased over K after
+// QK^T (two barriers per tile), per-wave P tile.
+// ============================================================================
+
+__global__ __launch_bounds__(256, 2) void attn_flash_fwd_kernel(
+    const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ mask,
+    ushort_t* __restrict__ out, float* __restrict__ m_io,
+    float* __restrict__ l_io, int B, int S, int h, float scale, float keep,
+    uint64_t salt, const unsigned long long* __restrict__ state) {
+  extern __shared__ __attribute__((aligned(16))) char lds[];
+  const int tid = threadIdx.x;
+  const int l = tid & 63;
+  const int w = tid >> 6;
+  const int nqb = (S + 127) / 128;
+  const int bh = blockIdx.x / nqb;
+  const int qb = blockIdx.x % nqb;
+  const int b = bh / h;
+  const int hh = bh % h;
+  const int ts = 3 * h * ATT_D;
+  const uint64_t seed = salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull;
+  const float inv_keep = 1.f / keep;
+  const unsigned keep16 = keep_to_16(keep);
+
+  const ushort_t* qbase = qkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
+  const ushort_t* kbase = qbase + (size_t)h * ATT_D;
+  const ushort_t* vbase = qbase + (size_t)2 * h * ATT_D;
+
+  const int lm = l & 15;
+  const int lg = l >> 4;
+  const int q0 = qb * 128;          // this block's first query row
+  const int qt0 = 2 * w;            // wave's first q tile within the block
+
+  // Q fragments for this wave's two q tiles (rows clamped)
+  bf16x8 aq_all[2][2];
+#pragma unroll
+  for (int qi = 0; qi < 2; ++qi) {
+    int qtok = q0 + (qt0 + qi) * 16 + lm;
+    if (qtok >= S) qtok = S - 1;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)
+      aq_all[qi][ks] = *(const bf16x8*)(qbase + (size_t)qtok * ts + ks * 32 + lg * 8);
+  }
+
+  // running state per (qi, r): max, sum; O accumulator
+  float mx[2][4], sm[2][4];
+  f32x4 oacc[2][4];
+#pragma unroll
+  for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      mx[qi][r] = -3.0e38f;
+      sm[qi][r] = 0.f;
+    }
+#pragma unroll
+  for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+    for (int dv = 0; dv < 4; ++dv) oacc[qi][dv] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int nkv = (S + 127) / 128;
+  for (int kv = 0; kv < nkv; ++kv) {
+    const int k0 = kv * 128;
+    const int ktiles = min(8, (S - k0 + 15) / 16);
+    // ---- stage K tile [128][64] via glds (source-side swizzle) ----
+    __syncthreads();  // previous tile's Vt reads complete before overwrite
+    for (int u = tid; u < 128 * 8; u += 256) {
+      const int tok = u >> 3;
+      int gtok = k0 + tok;
+      if (gtok >= S) gtok = S - 1;  // clamped; masked off via mval
+      const int c16s = (u & 7) ^ (tok & 7);
+      __builtin_amdgcn_global_load_lds(
+          (att_gas)(kbase + (size_t)gtok * ts + c16s * 8),
+          (att_las)lds_at(lds, K_OFF + u * 16), 16, 0, 0);
+    }
+    // additive mask values for this tile's columns
+    float mval[8];
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      const int col = k0 + kt * 16 + lm;
+      float mv = 0.f;
+      if (mask && col < S) mv = bf16_to_f32(mask[(size_t)b * S + col]);
+      mval[kt] = (col < S) ? mv : -3.0e38f;
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+#pragma clang loop unroll(disable)
+    for (int qi = 0; qi < 2; ++qi) {
+      const int qrow_base = q0 + (qt0 + qi) * 16;
+      if (qrow_base >= S) break;
+      f32x4 sacc[8];
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) sacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        if (kt >= ktiles) continue;
+        const int ktok = kt * 16 + lm;
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          bf16x8 bk = *(const bf16x8*)lds_at(
+              lds, swz(K_OFF + ktok * 128 + (ks * 32 + lg * 8) * 2, ktok, 7));
+          sacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aq_all[qi][ks], bk, sacc[kt], 0, 0, 0);
+        }
+      }
+      // online softmax update for this tile
+      float tmx[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) tmx[r] = -3.0e38f;
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        if (kt >= ktiles) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float zz = sacc[kt][r] * scale + mval[kt];
+          sacc[kt][r] = zz;
+          tmx[r] = fmaxf(tmx[r], zz);
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1)
+          tmx[r] = fmaxf(tmx[r], __shfl_xor(tmx[r], off, 64));
+        const float mn = fmaxf(mx[qi][r], tmx[r]);
+        const float f = __expf(mx[qi][r] - mn);  // exp(-inf - mn) -> 0 ok
+        mx[qi][r] = mn;
+        sm[qi][r] *= f;
+        // rescale O rows r of this qi (row index = lg*4+r of each 16x16)
+#pragma unroll
+        for (int dv = 0; dv < 4; ++dv) oacc[qi][dv][r] *= f;
+      }
+      float tsum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        if (kt >= ktiles) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float p = __expf(sacc[kt][r] - mx[qi][r]);
+          sacc[kt][r] = p;
+          tsum[r] += p;
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) tsum[r] += __shfl_xor(tsum[r], off, 64);
+        sm[qi][r] += tsum[r];
+      }
+      // dropout + write UNNORMALIZED P tile to per-wave LDS (linear-index
+      // RNG so the generic dropout kernels can regenerate the mask)
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        const int col = k0 + kt * 16 + lm;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = qrow_base + lg * 4 + r;
+          float p = (kt < ktiles) ? sacc[kt][r] : 0.f;
+          if (keep < 1.f && p != 0.f && row < S && col < S) {
+            const uint64_t idx = ((uint64_t)bh * S + row) * S + col;
+            p = rng_keep16(seed, idx, keep16) ? p * inv_keep : 0.f;
+          }
+          const int rl = qi * 16 + lg * 4 + r;
+          *(ushort_t*)lds_at(
+              lds, swz(P_OFF + w * 8192 + rl * 256 + (kt * 16 + lm) * 2, rl, 15)) =
+              f32_to_bf16((row < S && col < S) ? p : 0.f);
+        }
+      }
+    }
+
+    // ---- stage V^T over K's region, then accumulate PV ----
+    __syncthreads();  // K reads + P writes done
+    for (int u = tid; u < 128 * 8; u += 256) {
+      const int tok = u >> 3;
+      const int gtok = k0 + tok;
+      const int c16 = u & 7;
+      ushort8_t v;
+      if (gtok < S)
+        v = *(const ushort8_t*)(vbase + (size_t)gtok * ts + c16 * 8);
+      else
+        v = (ushort8_t)(ushort_t)0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int c = c16 * 8 + j;
+        *(ushort_t*)lds_at(lds, swz(VT_OFF + c * 256 + tok * 2, c, 15)) = v[j];
+      }
+    }
+    __syncthreads();
+#pragma clang loop unroll(disable)
+    for (int qi = 0; qi < 2; ++qi) {
+      if (q0 + (qt0 + qi) * 16 >= S) break;
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        const int rl = qi * 16 + lm;
+        bf16x8 ap = *(const bf16x8*)lds_at(
+            lds, swz(P_OFF + w * 8192 + rl * 256 + (ks * 32 + lg * 8) * 2, rl, 15));
+#pragma unroll
+        for (int dvt = 0; dvt < 4; ++dvt) {
+          const int dv = dvt * 16 + lm;
+          bf16x8 bv = *(const bf16x8*)lds_at(
+              lds, swz(VT_OFF + dv * 256 + (ks * 32 + lg * 8) * 2, dv, 15));
+          oacc[qi][dvt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, oacc[qi][dvt], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  // ---- epilogue: normalize by l, write out + stats ----
+#pragma unroll
+  for (int qi = 0; qi < 2; ++qi) {
+    const int qrow_base = q0 + (qt0 + qi) * 16;
+    if (qrow_base >= S) break;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qrow_base + lg * 4 + r;
+      const float inv = 1.f / sm[qi][r];
+      if (lm == 0 && row < S) {
+        m_io[(size_t)bh * S + row] = mx[qi][r];
+        l_io[(size_t)bh * S + row] = sm[qi][r];
+      }
+#pragma unroll
+      for (int dvt = 0; dvt < 4; ++dvt) {
+        const int dv = dvt * 16 + lm;
+        if (row < S)
+          out[(((size_t)b * S + row) * h + hh) * ATT_D + dv] =
+              f32_to_bf16(oacc[qi][dvt][r] * inv);
+      }
+    }
+  }
+}
+
+SKY_EXPORT int sky_attn_flash_fwd(uint64_t stream, uint64_t qkv,
+                                  uint64_t mask, uint64_t out, uint64_t m,
+                                  uint64_t lsum, int64_t B, int64_t S,
+                                  int64_t h, int64_t d, float scale,
+                                  float keep, uint64_t salt, uint64_t state) {
+  if (d != ATT_D) return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  const int nqb = (int)((S + 127) / 128);
+  dim3 grid((unsigned)(B * h * nqb));
+  hipLaunchKernelGGL(attn_flash_fwd_kernel, grid, dim3(256), ATT_LDS_BYTES, s,
+                     (const ushort_t*)qkv, (const ushort_t*)mask,
+                     (ushort_t*)out, (float*)m, (float*)lsum, (int)B, (int)S,
+                     (int)h, scale, keep, salt,
+                     (const unsigned long long*)state);
+  LAUNCH_CHECK();
+  return 0;
+}

@@ -64,6 +64,8 @@ def _register_signatures(lib):
         #                strm qkv  mask out  m    l    B    S    h    d  scale keep salt state
         "sky_attn_probs": [u64, u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, u64],
         #                  strm qkv mask m    l    p    pd   B    S    h    d  scale keep salt state
+        "sky_attn_flash_fwd": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, u64],
+        #                      strm qkv mask out  m    l    B    S    h    d scale keep salt state
         "sky_attn_bwd": [u64] * 9 + [i64, i64, i64, i64, f32, f32, u64, u64],
         # strm qkv dout mask m l pdT dsT dqkv | B S h d scale keep salt state
     }

@@ -490,3 +490,49 @@ def test_bias_gelu_odd_width_fallback():
     assert torch.allclose(y, yr, atol=1e-5, rtol=1e-5)
     assert torch.allclose(x.grad, xf.grad, atol=1e-5, rtol=1e-5)
     assert torch.allclose(b.grad, bf.grad, atol=1e-3, rtol=1e-3)
+
+
+@pytest.mark.parametrize("S", [192, 256, 512])
+def test_flash_attention_long_seq(S):
+    """Flash-style forward (online softmax over kv tiles) + decomposed
+    backward for S > 128, vs the eager fp32 reference."""
+    torch.manual_seed(20)
+    from skycomputing_amd.ops.functions import FlashAttentionFn
+
+    B, h, d = 2, 4, 64
+    scale = 1.0 / d ** 0.5
+    qkv = torch.randn(B, S, 3, h, d, dtype=torch.bfloat16, device="cuda",
+                      requires_grad=True)
+    mask = torch.zeros(B, 1, 1, S, dtype=torch.bfloat16, device="cuda")
+    mask[:, :, :, S - S // 5:] = -10000.0
+    out = FlashAttentionFn.apply(qkv, mask, scale, 0.0, False)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+
+    qf = qkv.detach().float().requires_grad_(True)
+    q = qf[:, :, 0].permute(0, 2, 1, 3)
+    k = qf[:, :, 1].permute(0, 2, 1, 3)
+    v = qf[:, :, 2].permute(0, 2, 1, 3)
+    ref = eager.attention_context(q, k, v, mask.float()).permute(0, 2, 1, 3)
+    ref.backward(dout.float())
+    assert torch.allclose(out.float(), ref, atol=6e-2, rtol=6e-2), (
+        S, (out.float() - ref).abs().max()
+    )
+    assert torch.allclose(qkv.grad.float(), qf.grad, atol=8e-2, rtol=8e-2), (
+        S, (qkv.grad.float() - qf.grad).abs().max()
+    )
+
+
+def test_flash_attention_dropout_trains():
+    torch.manual_seed(21)
+    from skycomputing_amd.ops.functions import FlashAttentionFn
+
+    qkv = torch.randn(2, 256, 3, 4, 64, dtype=torch.bfloat16, device="cuda",
+                      requires_grad=True)
+    out = FlashAttentionFn.apply(qkv, None, 0.125, 0.3, True)
+    assert torch.isfinite(out.float()).all()
+    out.sum().backward()
+    assert torch.isfinite(qkv.grad.float()).all()
+    # E[dropout out] ~ no-dropout out
+    out0 = FlashAttentionFn.apply(qkv.detach(), None, 0.125, 0.0, False)
+    assert (out.float().mean() - out0.float().mean()).abs() < 0.05
