@@ -128,8 +128,11 @@ def attention(qkv, mask, dropout_p: float = 0.0, training: bool = False):
         if S <= 128:
             out = FusedAttentionFn.apply(qkv, mask, scale, dropout_p, training)
             return out.reshape(B, S, h * d)
-        if S <= 4096:
-            # flash-style forward (online softmax); decomposed backward
+        if S <= 4096 and os.environ.get("SKY_FLASH_ATTN") == "1":
+            # flash-style forward (online softmax, no S x S tensor in
+            # forward). Currently a MEMORY option: the decomposed path
+            # below measures faster at these shapes (profiles/r01_notes.md),
+            # so it stays the default.
             out = FlashAttentionFn.apply(qkv, mask, scale, dropout_p, training)
             return out.reshape(B, S, h * d)
     q = qkv[:, :, 0].permute(0, 2, 1, 3)
