@@ -131,11 +131,18 @@ if HAVE_HYP:
             for r in range(W)
         ]
         a = Allocator(flops, mem, workers)
-        try:
-            pe = a.even_allocate()
-            ce = a.plan_cost(pe)
-        except AllocationError:
-            ce = None
+
+        def _mem_ok(plan):
+            by_rank = {w["rank"]: w for w in a.workers}
+            return all(
+                sum(mem[s:e]) <= by_rank[r]["avai_mem"] * (1 + 1e-9)
+                for r, (s, e) in zip(plan.stage_ranks, plan.ranges)
+            )
+
+        # even_allocate ignores memory (reference parity); its cost is only a
+        # valid comparison point when the even plan happens to fit.
+        pe = a.even_allocate()
+        ce = a.plan_cost(pe) if _mem_ok(pe) else None
         try:
             pd = a.dynamic_allocate()
             _check_plan(pd, L)
