@@ -47,9 +47,10 @@ class Allocator:
         e.g. Body->Tail cuts whose hop carries the 4x intermediate tensor,
         SURVEY.md §2c C4). It enters ``_chunk_time`` and therefore the
         dynamic allocator's objective and ``plan_cost``; the exact optimal
-        solver currently optimizes pure compute time (its binary-search
-        extension step needs monotone chunk cost) — run dynamic_allocate
-        on top of an optimal plan when cut costs matter."""
+        solver's binary-search extension step needs monotone chunk cost, so
+        it optimizes pure compute time and then, when cut costs are
+        configured, runs a boundary-shift local search (``refine_plan``)
+        over its exact device order to move cuts off expensive edges."""
         self.L = len(layer_flops)
         assert len(layer_mem) == self.L
         self.flops = [float(f) for f in layer_flops]
@@ -120,25 +121,30 @@ class Allocator:
             bounds.append(bounds[-1] + base + (1 if i < rem else 0))
         ws = self.workers
         bounds = self._repair_memory(bounds, ws)
-        # balance by time
+        bounds = self._balance_bounds(bounds, ws, break_iter)
+        return self._plan_from_bounds(bounds, [w["rank"] for w in ws])
+
+    def _balance_bounds(self, bounds: list[int], ws: list[dict], break_iter: int) -> list[int]:
+        """Iterative single-layer boundary shifts toward minimizing the
+        bottleneck chunk (under the full ``_chunk_time`` objective, so cut
+        costs count when configured)."""
+        n = len(ws)
         for _ in range(break_iter):
-            times = [self._chunk_time(ws[i], bounds[i], bounds[i + 1]) for i in range(self.W)]
-            worst = max(range(self.W), key=lambda i: times[i])
+            times = [self._chunk_time(ws[i], bounds[i], bounds[i + 1]) for i in range(n)]
+            worst = max(range(n), key=lambda i: times[i])
             improved = False
             # try shedding one layer from the bottleneck to either neighbor
             for di in (-1, 1):
                 j = worst + di
-                if not (0 <= j < self.W):
+                if not (0 <= j < n):
                     continue
                 nb = list(bounds)
+                if nb[worst + 1] - nb[worst] <= 0:
+                    continue
                 if di == -1:
                     # give the first layer of `worst` to the left neighbor
-                    if nb[worst + 1] - nb[worst] <= 0:
-                        continue
                     nb[worst] += 1
                 else:
-                    if nb[worst + 1] - nb[worst] <= 0:
-                        continue
                     nb[worst + 1] -= 1
                 if not self._chunk_mem_ok(ws[j], nb[j], nb[j + 1]):
                     continue
@@ -152,7 +158,19 @@ class Allocator:
                     break
             if not improved:
                 break
-        return self._plan_from_bounds(bounds, [w["rank"] for w in ws])
+        return bounds
+
+    def refine_plan(self, plan: PartitionPlan, break_iter: int = 1000) -> PartitionPlan:
+        """Boundary-shift local search over a plan's FIXED stage order —
+        improves cut placement under the full objective (incl. comm costs)
+        without changing which devices participate or their order. Never
+        returns a worse plan than the input."""
+        by_rank = {w["rank"]: w for w in self.workers}
+        ws = [by_rank[r] for r in plan.stage_ranks]
+        bounds = [plan.ranges[0][0]] + [b for _a, b in plan.ranges]
+        bounds = self._balance_bounds(bounds, ws, break_iter)
+        refined = self._plan_from_bounds(bounds, [w["rank"] for w in ws])
+        return refined if self.plan_cost(refined) <= self.plan_cost(plan) else plan
 
     def _repair_memory(self, bounds: list[int], ws: list[dict]) -> list[int]:
         """Shift layers off over-memory chunks (reference:
@@ -267,7 +285,10 @@ class Allocator:
             raise AllocationError("optimal allocation reconstruction failed")
         stage_ranks = [c[0] for c in chunks]
         ranges = [(c[1], c[2]) for c in chunks]
-        return PartitionPlan(stage_ranks=stage_ranks, ranges=ranges)
+        plan = PartitionPlan(stage_ranks=stage_ranks, ranges=ranges)
+        if self.boundary_cost is not None and self.comm_weight > 0:
+            plan = self.refine_plan(plan)
+        return plan
 
     def allocate(self, mode: str = "optimal") -> PartitionPlan:
         if mode == "even":

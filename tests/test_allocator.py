@@ -183,6 +183,41 @@ def test_comm_aware_dynamic_avoids_expensive_cut():
     assert a1.plan_cost(p1) < a1.plan_cost(p0)
 
 
+def test_comm_aware_optimal_avoids_expensive_cut():
+    """optimal_allocate refines its exact compute-optimal order with a
+    boundary local search when cut costs are configured."""
+    L, W = 12, 2
+    flops = [1.0] * L
+    mem = [0.0] * L
+    workers = [dict(rank=r, time=1.0, avai_mem=100.0) for r in range(W)]
+    cost = [0.0] * (L - 1)
+    cost[5] = 1000.0  # the pure-compute optimum would cut exactly here
+    a0 = Allocator(flops, mem, workers)
+    a1 = Allocator(flops, mem, workers, boundary_cost=cost, comm_weight=1.0)
+    p0 = a0.optimal_allocate()
+    p1 = a1.optimal_allocate()
+    assert p0.ranges[0][1] == 6
+    assert p1.ranges[0][1] != 6, p1
+    assert a1.plan_cost(p1) < a1.plan_cost(p0)
+    _check_plan(p1, L)
+
+
+def test_refine_plan_never_worse():
+    rng = random.Random(7)
+    for _ in range(30):
+        L = rng.randint(4, 30)
+        W = rng.randint(2, 5)
+        flops = [rng.uniform(0.1, 2.0) for _ in range(L)]
+        mem = [rng.uniform(0.5, 1.5) for _ in range(L)]
+        cost = [rng.uniform(0.0, 3.0) for _ in range(L - 1)]
+        workers = [dict(rank=r, time=rng.uniform(1.0, 4.0), avai_mem=sum(mem)) for r in range(W)]
+        a = Allocator(flops, mem, workers, boundary_cost=cost, comm_weight=rng.uniform(0.0, 1.0))
+        p = a.even_allocate()
+        r = a.refine_plan(p)
+        assert a.plan_cost(r) <= a.plan_cost(p) + 1e-12
+        _check_plan(r, L)
+
+
 def test_bert_boundary_payloads_shape():
     from skycomputing_amd.dynamics import bert_boundary_payloads
     from skycomputing_amd.models import bert_pipeline_config
