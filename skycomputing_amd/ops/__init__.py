@@ -154,8 +154,14 @@ def linear(x, weight, bias=None):
 def linear_act(x, weight, bias, act: str = "gelu"):
     """Linear + bias + activation (the reference's LinearActivation).
 
-    GPU: hipBLASLt GEMM + fused HIP bias+act epilogue kernel."""
+    GPU: ONE hipBLASLt GEMM with the GELU_AUX_BIAS epilogue (bias + gelu +
+    pre-activation save fused into the GEMM); SKY_NO_HBLT=1 falls back to
+    GEMM + the standalone HIP bias+gelu kernel."""
     if _use_hip(x) and act == "gelu" and bias is not None:
+        from .functions import LinearGeluFn, _use_hblt
+
+        if _use_hblt() and weight.shape[0] % 8 == 0:
+            return LinearGeluFn.apply(x, weight, bias)
         y = torch.nn.functional.linear(x, weight)
         return bias_gelu(y, bias)
     return eager.linear_act(x, weight, bias, act)

@@ -59,7 +59,8 @@ def build(force: bool = False, verbose: bool = True) -> str:
             print("[skyhip]", " ".join(cmd))
         subprocess.run(cmd, check=True)
         objs.append(obj)
-    cmd = [HIPCC, f"--offload-arch={ARCH}", "-shared", "-fPIC", *objs, "-o", LIB]
+    cmd = [HIPCC, f"--offload-arch={ARCH}", "-shared", "-fPIC", *objs,
+           "-L/opt/rocm/lib", "-lhipblaslt", "-o", LIB]
     if verbose:
         print("[skyhip]", " ".join(cmd))
     subprocess.run(cmd, check=True)

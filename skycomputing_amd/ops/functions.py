@@ -484,10 +484,16 @@ def colsum(src: torch.Tensor, out_dtype=None) -> torch.Tensor:
     return out
 
 
+def _use_hblt() -> bool:
+    return os.environ.get("SKY_NO_HBLT") != "1"
+
+
 class LinearBiasFn(torch.autograd.Function):
-    """Linear + bias with a custom backward: dgrad/wgrad stay hipBLASLt
-    GEMMs, but dbias uses the HIP column reduction instead of torch's
-    generic reduce_kernel (a measured hot spot — profiles/r01_notes.md)."""
+    """Linear + bias with a custom backward: dgrad stays a hipBLASLt GEMM;
+    wgrad is a direct hipBLASLt call whose BGRADB epilogue produces dbias
+    inside the GEMM (ops/hip/hblt.hip), removing the separate column
+    reduction per linear (a measured hot spot — profiles/r01_notes.md).
+    SKY_NO_HBLT=1 falls back to torch wgrad + HIP colsum."""
 
     @staticmethod
     def forward(ctx, x, weight, bias):
@@ -506,8 +512,74 @@ class LinearBiasFn(torch.autograd.Function):
         if not dy2.is_contiguous():
             dy2 = dy2.contiguous()
         dx = dy2.mm(weight).view(ctx.xshape)
+        if ctx.has_bias and _use_hblt():
+            lib = hiplib.require()
+            M, K = x2.shape
+            N = weight.shape[0]
+            dw = torch.empty_like(weight)
+            db = torch.empty(N, dtype=weight.dtype, device=weight.device)
+            check(
+                lib.sky_hblt_wgrad_bgrad(
+                    _stream(), ptr(x2), ptr(dy2), ptr(dw), ptr(db), M, N, K, _dt(x2)
+                ),
+                "sky_hblt_wgrad_bgrad",
+            )
+            return dx, dw, db
         dw = dy2.t().mm(x2)
         db = colsum(dy2, weight.dtype) if ctx.has_bias else None
+        return dx, dw, db
+
+
+class LinearGeluFn(torch.autograd.Function):
+    """Fused linear + bias + GELU via the hipBLASLt GELU_AUX_BIAS epilogue
+    (the reference's LinearActivation, scaelum/model/bert_layers.py:60-108):
+    the forward is ONE GEMM that also writes the pre-activation `aux`; the
+    backward reuses the bias_gelu kernel with aux as the pre-activation
+    (dpre + fused dbias), then plain dgrad/wgrad GEMMs. hipBLASLt's GELU is
+    the tanh approximation; the erf-derivative backward differs by less
+    than bf16 rounding (tests/test_ops_gpu.py)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        lib = hiplib.require()
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1]).contiguous()
+        M, K = x2.shape
+        N = weight.shape[0]
+        y = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
+        aux = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
+        check(
+            lib.sky_hblt_linear_gelu_aux(
+                _stream(), ptr(x2), ptr(weight), ptr(bias), ptr(y), ptr(aux),
+                M, N, K, _dt(x2)
+            ),
+            "sky_hblt_linear_gelu_aux",
+        )
+        ctx.save_for_backward(x2, weight, aux)
+        ctx.xshape = xs
+        return y.view(*xs[:-1], N)
+
+    @staticmethod
+    def backward(ctx, dy):
+        lib = hiplib.require()
+        x2, weight, aux = ctx.saved_tensors
+        N = weight.shape[0]
+        dy2 = dy.reshape(-1, N)
+        if not dy2.is_contiguous():
+            dy2 = dy2.contiguous()
+        rows = dy2.shape[0]
+        dpre = torch.empty_like(dy2)
+        db = torch.empty(N, dtype=weight.dtype, device=weight.device)
+        scratch = _red_scratch(N, 1, dy2.device)
+        check(
+            lib.sky_bias_gelu_bwd(
+                _stream(), ptr(dy2), ptr(aux), 0, ptr(dpre), ptr(db),
+                ptr(scratch), rows, N, _dt(dy2)
+            ),
+            "sky_bias_gelu_bwd",
+        )
+        dx = dpre.mm(weight).view(ctx.xshape)
+        dw = dpre.t().mm(x2)
         return dx, dw, db
 
 

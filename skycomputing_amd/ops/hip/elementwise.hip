@@ -68,8 +68,10 @@ SKY_EXPORT int sky_bias_gelu_fwd(uint64_t stream, uint64_t x, uint64_t b,
 
 // ---------------- bias_gelu backward ----------------
 // dx = dy * gelu'(x+b), then db[c] = sum_r dx[r,c] via column reduction.
+// With b == nullptr, x is already the pre-activation (the hipBLASLt
+// GELU_AUX_BIAS epilogue saved it) and the add is skipped.
 
-template <int DT>
+template <int DT, bool HAS_B>
 __global__ __launch_bounds__(256) void bias_gelu_bwd_dx_vec_kernel(
     const void* __restrict__ dy, const void* __restrict__ x,
     const void* __restrict__ b, void* __restrict__ dx, int64_t n8,
@@ -78,22 +80,22 @@ __global__ __launch_bounds__(256) void bias_gelu_bwd_dx_vec_kernel(
        i8 += (int64_t)gridDim.x * 256) {
     float v[8], bv[8], d[8];
     Vec8<DT>::load(x, i8, v);
-    Vec8<DT>::load(b, i8 % cols8, bv);
+    if (HAS_B) Vec8<DT>::load(b, i8 % cols8, bv);
     Vec8<DT>::load(dy, i8, d);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) d[j] *= gelu_grad_f(v[j] + bv[j]);
+    for (int j = 0; j < 8; ++j) d[j] *= gelu_grad_f(HAS_B ? v[j] + bv[j] : v[j]);
     Vec8<DT>::store(dx, i8, d);
   }
 }
 
-template <int DT>
+template <int DT, bool HAS_B>
 __global__ __launch_bounds__(256) void bias_gelu_bwd_dx_kernel(
     const void* __restrict__ dy, const void* __restrict__ x,
     const void* __restrict__ b, void* __restrict__ dx, int64_t n,
     int64_t cols) {
   for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * 256) {
-    float z = load_elem<DT>(x, i) + load_elem<DT>(b, i % cols);
+    float z = load_elem<DT>(x, i) + (HAS_B ? load_elem<DT>(b, i % cols) : 0.f);
     store_elem<DT>(dx, i, load_elem<DT>(dy, i) * gelu_grad_f(z));
   }
 }
@@ -217,25 +219,34 @@ SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
                                  int dt) {
   hipStream_t s = (hipStream_t)stream;
   int64_t n = rows * cols;
+  const bool hb = b != 0;
   if (cols % 8 == 0) {
     int64_t n8 = n / 8;
     unsigned grid = (unsigned)((n8 + 255) / 256);
     if (grid > 2048u) grid = 2048u;
+    auto launch_vec = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(grid), dim3(256), 0, s, (const void*)dy,
+                         (const void*)x, (const void*)b, (void*)dx, n8, cols / 8);
+    };
     if (dt == DT_F32)
-      hipLaunchKernelGGL((bias_gelu_bwd_dx_vec_kernel<DT_F32>), dim3(grid), dim3(256), 0, s,
-                         (const void*)dy, (const void*)x, (const void*)b, (void*)dx, n8, cols / 8);
+      hb ? launch_vec(bias_gelu_bwd_dx_vec_kernel<DT_F32, true>)
+         : launch_vec(bias_gelu_bwd_dx_vec_kernel<DT_F32, false>);
     else
-      hipLaunchKernelGGL((bias_gelu_bwd_dx_vec_kernel<DT_BF16>), dim3(grid), dim3(256), 0, s,
-                         (const void*)dy, (const void*)x, (const void*)b, (void*)dx, n8, cols / 8);
+      hb ? launch_vec(bias_gelu_bwd_dx_vec_kernel<DT_BF16, true>)
+         : launch_vec(bias_gelu_bwd_dx_vec_kernel<DT_BF16, false>);
   } else {
     unsigned grid = (unsigned)((n + 255) / 256);
     if (grid > 2048u) grid = 2048u;
+    auto launch_sc = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(grid), dim3(256), 0, s, (const void*)dy,
+                         (const void*)x, (const void*)b, (void*)dx, n, cols);
+    };
     if (dt == DT_F32)
-      hipLaunchKernelGGL((bias_gelu_bwd_dx_kernel<DT_F32>), dim3(grid), dim3(256), 0, s,
-                         (const void*)dy, (const void*)x, (const void*)b, (void*)dx, n, cols);
+      hb ? launch_sc(bias_gelu_bwd_dx_kernel<DT_F32, true>)
+         : launch_sc(bias_gelu_bwd_dx_kernel<DT_F32, false>);
     else
-      hipLaunchKernelGGL((bias_gelu_bwd_dx_kernel<DT_BF16>), dim3(grid), dim3(256), 0, s,
-                         (const void*)dy, (const void*)x, (const void*)b, (void*)dx, n, cols);
+      hb ? launch_sc(bias_gelu_bwd_dx_kernel<DT_BF16, true>)
+         : launch_sc(bias_gelu_bwd_dx_kernel<DT_BF16, false>);
   }
   if (dt == DT_F32)
     launch_colsum<DT_F32>(s, (const void*)dx, (void*)db, (float*)scratch, rows, cols, dt);
