@@ -160,7 +160,10 @@ def linear_act(x, weight, bias, act: str = "gelu"):
     if _use_hip(x) and act == "gelu" and bias is not None:
         from .functions import LinearGeluFn, _use_hblt
 
-        if _use_hblt() and weight.shape[0] % 8 == 0:
+        # GELU_AUX_BIAS epilogue: opt-in until the bf16-aux support on this
+        # hipBLASLt build is confirmed (tools/hblt_probe.py)
+        if (_use_hblt() and os.environ.get("SKY_HBLT_GELU") == "1"
+                and weight.shape[0] % 8 == 0):
             return LinearGeluFn.apply(x, weight, bias)
         y = torch.nn.functional.linear(x, weight)
         return bias_gelu(y, bias)

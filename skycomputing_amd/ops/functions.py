@@ -551,7 +551,7 @@ class LinearGeluFn(torch.autograd.Function):
         check(
             lib.sky_hblt_linear_gelu_aux(
                 _stream(), ptr(x2), ptr(weight), ptr(bias), ptr(y), ptr(aux),
-                M, N, K, _dt(x2)
+                M, N, K, _dt(x2), _dt(aux)
             ),
             "sky_hblt_linear_gelu_aux",
         )

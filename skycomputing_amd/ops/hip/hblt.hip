@@ -45,8 +45,10 @@ struct PlanKey {
   int kind;  // 0 = wgrad+bgradb, 1 = linear+gelu_aux+bias
   int64_t M, N, K;
   int dt;
+  int aux_dt;  // DT_* of the saved pre-activation (kind 1 only)
   bool operator<(const PlanKey& o) const {
-    return std::tie(kind, M, N, K, dt) < std::tie(o.kind, o.M, o.N, o.K, o.dt);
+    return std::tie(kind, M, N, K, dt, aux_dt) <
+           std::tie(o.kind, o.M, o.N, o.K, o.dt, o.aux_dt);
   }
 };
 
@@ -54,6 +56,9 @@ struct Plan {
   hipblasLtMatmulDesc_t desc = nullptr;
   hipblasLtMatrixLayout_t la = nullptr, lb = nullptr, ld = nullptr;
   hipblasLtMatmulAlgo_t algo;
+  hipblasLtMatmulAlgo_t candidates[16];
+  int n_candidates = 0;
+  bool tuned = false;
   size_t workspace = 0;
 };
 
@@ -85,9 +90,9 @@ hipDataType to_hip_dt(int dt) { return dt == DT_BF16 ? HIP_R_16BF : HIP_R_32F; }
 //   kind 1 (fwd):   D(N,M) colmaj [= Y row-major [M,N]] = op(A)·B,
 //     A = W (colmaj (K,N), op T, lda K), B = X (colmaj (K,M), op N, ldb K);
 //     bias length = D rows = N, broadcast over columns (tokens).
-int get_plan(int kind, int64_t M, int64_t N, int64_t K, int dt,
+int get_plan(int kind, int64_t M, int64_t N, int64_t K, int dt, int aux_dt,
              const void* bias_or_db, const void* aux, Plan** out) {
-  PlanKey key{kind, M, N, K, dt};
+  PlanKey key{kind, M, N, K, dt, aux_dt};
   auto it = g_plans.find(key);
   if (it != g_plans.end()) {
     *out = &it->second;
@@ -120,6 +125,8 @@ int get_plan(int kind, int64_t M, int64_t N, int64_t K, int dt,
     HBLT_CHECK(hipblasLtMatmulDescSetAttribute(p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_POINTER, &aux, sizeof(aux)));
     int64_t aux_ld = N;
     HBLT_CHECK(hipblasLtMatmulDescSetAttribute(p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &aux_ld, sizeof(aux_ld)));
+    int32_t adt = (int32_t)to_hip_dt(aux_dt);
+    HBLT_CHECK(hipblasLtMatmulDescSetAttribute(p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_DATA_TYPE, &adt, sizeof(adt)));
     HBLT_CHECK(hipblasLtMatrixLayoutCreate(&p.la, hdt, K, N, K));   // W
     HBLT_CHECK(hipblasLtMatrixLayoutCreate(&p.lb, hdt, K, M, K));   // X
     HBLT_CHECK(hipblasLtMatrixLayoutCreate(&p.ld, hdt, N, M, N));   // Y
@@ -128,28 +135,76 @@ int get_plan(int kind, int64_t M, int64_t N, int64_t K, int dt,
   HBLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
   uint64_t ws = kMaxWorkspace;
   HBLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
-  hipblasLtMatmulHeuristicResult_t results[8];
+  hipblasLtMatmulHeuristicResult_t results[16];
   int n_results = 0;
   hipblasStatus_t hs = hipblasLtMatmulAlgoGetHeuristic(
-      g_handle, p.desc, p.la, p.lb, p.ld, p.ld, pref, 8, results, &n_results);
+      g_handle, p.desc, p.la, p.lb, p.ld, p.ld, pref, 16, results, &n_results);
   (void)hipblasLtMatmulPreferenceDestroy(pref);
   if (hs != HIPBLAS_STATUS_SUCCESS) return (int)hs;
   if (n_results == 0) return (int)HIPBLAS_STATUS_NOT_SUPPORTED;
   p.algo = results[0].algo;
   p.workspace = results[0].workspaceSize;
+  for (int i = 0; i < n_results; ++i)
+    if (results[i].workspaceSize > p.workspace) p.workspace = results[i].workspaceSize;
   int rc = ensure_workspace(p.workspace);
   if (rc) return rc;
+  p.n_candidates = n_results;
+  for (int i = 0; i < n_results && i < 16; ++i) p.candidates[i] = results[i].algo;
   auto ins = g_plans.emplace(key, p);
   *out = &ins.first->second;
   return 0;
 }
 
+// Measured algo selection: the heuristic's first suggestion is often NOT
+// the fastest once an epilogue is attached (measured +2 ms/step for the
+// BGRADB wgrad on the 160-layer bench when trusting results[0]). Time
+// every candidate on the real buffers and keep the winner. Runs once per
+// shape during the eager warmup steps; skipped (first candidate kept) if
+// the stream is capturing so plan misses can never corrupt a hipGraph.
+int tune_plan(Plan* p, hipStream_t stream, const void* A, const void* B,
+              void* D, float alpha, float beta) {
+  if (p->tuned || p->n_candidates <= 1) {
+    p->tuned = true;
+    return 0;
+  }
+  hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
+  (void)hipStreamIsCapturing(stream, &cap);
+  if (cap != hipStreamCaptureStatusNone) return 0;  // tune on a later call
+  hipEvent_t t0, t1;
+  if (hipEventCreate(&t0) != hipSuccess) return 0;
+  if (hipEventCreate(&t1) != hipSuccess) { (void)hipEventDestroy(t0); return 0; }
+  float best = 1e30f;
+  int best_i = 0;
+  for (int i = 0; i < p->n_candidates; ++i) {
+    hipblasStatus_t s = hipblasLtMatmul(g_handle, p->desc, &alpha, A, p->la,
+                                        B, p->lb, &beta, D, p->ld, D, p->ld,
+                                        &p->candidates[i], g_workspace,
+                                        g_workspace_bytes, stream);
+    if (s != HIPBLAS_STATUS_SUCCESS) continue;  // warm-up / validity check
+    (void)hipEventRecord(t0, stream);
+    for (int r = 0; r < 3; ++r)
+      (void)hipblasLtMatmul(g_handle, p->desc, &alpha, A, p->la, B, p->lb,
+                            &beta, D, p->ld, D, p->ld, &p->candidates[i],
+                            g_workspace, g_workspace_bytes, stream);
+    (void)hipEventRecord(t1, stream);
+    if (hipEventSynchronize(t1) != hipSuccess) continue;
+    float ms = 0.f;
+    (void)hipEventElapsedTime(&ms, t0, t1);
+    if (ms < best) { best = ms; best_i = i; }
+  }
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  p->algo = p->candidates[best_i];
+  p->tuned = true;
+  return 0;
+}
+
 int run(int kind, hipStream_t stream, const void* A, const void* B,
         const void* bias_or_db, const void* aux, void* D,
-        int64_t M, int64_t N, int64_t K, int dt) {
+        int64_t M, int64_t N, int64_t K, int dt, int aux_dt) {
   std::lock_guard<std::mutex> lock(g_mu);
   Plan* p = nullptr;
-  int rc = get_plan(kind, M, N, K, dt, bias_or_db, aux, &p);
+  int rc = get_plan(kind, M, N, K, dt, aux_dt, bias_or_db, aux, &p);
   if (rc) return rc;
   // the bias / aux pointers live in the cached desc: refresh per call
   HBLT_CHECK(hipblasLtMatmulDescSetAttribute(p->desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias_or_db, sizeof(bias_or_db)));
@@ -157,6 +212,7 @@ int run(int kind, hipStream_t stream, const void* A, const void* B,
     HBLT_CHECK(hipblasLtMatmulDescSetAttribute(p->desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_POINTER, &aux, sizeof(aux)));
   }
   float alpha = 1.0f, beta = 0.0f;
+  if (!p->tuned) tune_plan(p, stream, A, B, D, alpha, beta);
   HBLT_CHECK(hipblasLtMatmul(g_handle, p->desc, &alpha, A, p->la, B, p->lb,
                              &beta, D, p->ld, D, p->ld, &p->algo, g_workspace,
                              p->workspace, stream));
@@ -169,13 +225,15 @@ int run(int kind, hipStream_t stream, const void* A, const void* B,
 SKY_EXPORT int sky_hblt_wgrad_bgrad(hipStream_t stream, const void* x,
                                     const void* dy, void* dw, void* db,
                                     int64_t M, int64_t N, int64_t K, int dt) {
-  return run(0, stream, x, dy, db, nullptr, dw, M, N, K, dt);
+  return run(0, stream, x, dy, db, nullptr, dw, M, N, K, dt, dt);
 }
 
-// Y[M,N] = gelu(X[M,K] @ W[N,K]^T + bias[N]); aux[M,N] = pre-activation.
+// Y[M,N] = gelu(X[M,K] @ W[N,K]^T + bias[N]); aux[M,N] = pre-activation
+// (dtype aux_dt; hipBLASLt may only support fp32 aux for some D types).
 SKY_EXPORT int sky_hblt_linear_gelu_aux(hipStream_t stream, const void* x,
                                         const void* w, const void* bias,
                                         void* y, void* aux, int64_t M,
-                                        int64_t N, int64_t K, int dt) {
-  return run(1, stream, w, x, bias, aux, y, M, N, K, dt);
+                                        int64_t N, int64_t K, int dt,
+                                        int aux_dt) {
+  return run(1, stream, w, x, bias, aux, y, M, N, K, dt, aux_dt);
 }

@@ -19,6 +19,43 @@ struct SgdDesc {
   int64_t pad;
 };
 
+// Every byte in the SGD step is touch-once (grads, masters, params are not
+// re-read until the next iteration), so all vector I/O is NON-TEMPORAL:
+// streaming loads/stores bypass L2 retention and avoid write-allocate
+// pollution, which is worth real bandwidth on an 8 TB/s HBM3E part.
+template <int DT>
+DEV void nt_load8(const void* p, int64_t i8, float f[8]);
+template <>
+DEV void nt_load8<DT_BF16>(const void* p, int64_t i8, float f[8]) {
+  ushort8_t v = __builtin_nontemporal_load((const ushort8_t*)p + i8);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) f[j] = bf16_to_f32(v[j]);
+}
+template <>
+DEV void nt_load8<DT_F32>(const void* p, int64_t i8, float f[8]) {
+  float4_t a = __builtin_nontemporal_load((const float4_t*)p + i8 * 2);
+  float4_t b = __builtin_nontemporal_load((const float4_t*)p + i8 * 2 + 1);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) { f[j] = a[j]; f[4 + j] = b[j]; }
+}
+template <int DT>
+DEV void nt_store8(void* p, int64_t i8, const float f[8]);
+template <>
+DEV void nt_store8<DT_BF16>(void* p, int64_t i8, const float f[8]) {
+  ushort8_t v;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) v[j] = f32_to_bf16(f[j]);
+  __builtin_nontemporal_store(v, (ushort8_t*)p + i8);
+}
+template <>
+DEV void nt_store8<DT_F32>(void* p, int64_t i8, const float f[8]) {
+  float4_t a, b;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) { a[j] = f[j]; b[j] = f[4 + j]; }
+  __builtin_nontemporal_store(a, (float4_t*)p + i8 * 2);
+  __builtin_nontemporal_store(b, (float4_t*)p + i8 * 2 + 1);
+}
+
 template <int DT, int BLOCK, bool MASTER, bool MOM>
 __global__ __launch_bounds__(BLOCK) void sgd_kernel(
     const SgdDesc* __restrict__ descs, float lr, float momentum, float wd) {
@@ -28,27 +65,33 @@ __global__ __launch_bounds__(BLOCK) void sgd_kernel(
   float* master = (float*)d.m;
   float* mbuf = (float*)d.mom;
   const int64_t n8 = d.n / 8;
-  // vectorized body (slab bases are 16B-aligned; tail handled below)
-  for (int64_t i8 = threadIdx.x; i8 < n8; i8 += BLOCK) {
-    float gv[8], pv[8];
-    Vec8<DT>::load(g, i8, gv);
-    if (MASTER)
-      Vec8<DT_F32>::load(master, i8, pv);
-    else
-      Vec8<DT>::load(p, i8, pv);
+  // vectorized body (slab bases are 16B-aligned; tail handled below);
+  // 2x unrolled for memory-level parallelism
+  for (int64_t i8 = threadIdx.x; i8 < n8; i8 += 2 * BLOCK) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float gg = gv[j];
-      if (wd != 0.f) gg += wd * pv[j];
-      if (MOM) {
-        float b = mbuf[i8 * 8 + j] * momentum + gg;
-        mbuf[i8 * 8 + j] = b;
-        gg = b;
+    for (int u = 0; u < 2; ++u) {
+      const int64_t k8 = i8 + u * BLOCK;
+      if (k8 >= n8) break;
+      float gv[8], pv[8];
+      nt_load8<DT>(g, k8, gv);
+      if (MASTER)
+        nt_load8<DT_F32>(master, k8, pv);
+      else
+        nt_load8<DT>(p, k8, pv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float gg = gv[j];
+        if (wd != 0.f) gg += wd * pv[j];
+        if (MOM) {
+          float b = mbuf[k8 * 8 + j] * momentum + gg;
+          mbuf[k8 * 8 + j] = b;
+          gg = b;
+        }
+        pv[j] -= lr * gg;
       }
-      pv[j] -= lr * gg;
+      if (MASTER) nt_store8<DT_F32>(master, k8, pv);
+      nt_store8<DT>(p, k8, pv);
     }
-    if (MASTER) Vec8<DT_F32>::store(master, i8, pv);
-    Vec8<DT>::store(p, i8, pv);
   }
   for (int64_t i = n8 * 8 + threadIdx.x; i < d.n; i += BLOCK) {
     float gv = load_elem<DT>(g, i);

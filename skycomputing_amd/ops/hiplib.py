@@ -56,7 +56,7 @@ def _register_signatures(lib):
         "sky_rng_tick": [u64, u64],  # strm, state ptr
         "sky_colsum": [u64, u64, u64, u64, i64, i64, i32, i32],  # strm src out scratch rows cols dt dtout
         "sky_hblt_wgrad_bgrad": [u64, u64, u64, u64, u64, i64, i64, i64, i32],  # strm x dy dw db M N K dt
-        "sky_hblt_linear_gelu_aux": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i32],  # strm x w bias y aux M N K dt
+        "sky_hblt_linear_gelu_aux": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i32, i32],  # strm x w bias y aux M N K dt aux_dt
         "sky_mfma_probe": [u64, u64, u64, u64],  # strm A B D
         "sky_glds_probe": [u64, u64, u64, i32],
         "sky_gemm": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, i64, i64, i32, i32, i32, i32, i32],
