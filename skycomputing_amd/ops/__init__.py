@@ -160,10 +160,12 @@ def linear_act(x, weight, bias, act: str = "gelu"):
     if _use_hip(x) and act == "gelu" and bias is not None:
         from .functions import LinearGeluFn, _use_hblt
 
-        # GELU_AUX_BIAS epilogue: opt-in until the bf16-aux support on this
-        # hipBLASLt build is confirmed (tools/hblt_probe.py)
+        # GELU_AUX_BIAS epilogue: this hipBLASLt build only supports it for
+        # fp32 D (bf16 returns HIPBLAS_STATUS_NOT_SUPPORTED — probed on HW,
+        # tools/hblt_probe.py), so the fused linear+gelu is fp32-only and
+        # opt-in; the bf16 hot path keeps GEMM + the HIP bias+gelu kernel.
         if (_use_hblt() and os.environ.get("SKY_HBLT_GELU") == "1"
-                and weight.shape[0] % 8 == 0):
+                and x.dtype == torch.float32 and weight.shape[0] % 8 == 0):
             return LinearGeluFn.apply(x, weight, bias)
         y = torch.nn.functional.linear(x, weight)
         return bias_gelu(y, bias)

@@ -579,8 +579,17 @@ __global__ __launch_bounds__(256, 2) void attn_bwd1_kernel(
       }
       if (col < S) {
         const size_t o = ((size_t)bh * S + col) * S + qtok_base + lg * 4;
-        *(ushort4_t*)(dsT + o) = ds4;
-        *(ushort4_t*)(pdT + o) = pd4;
+        if (qtok_base + lg * 4 + 3 < S) {  // whole 4-pack in range
+          *(ushort4_t*)(dsT + o) = ds4;
+          *(ushort4_t*)(pdT + o) = pd4;
+        } else {  // edge tile of a non-multiple-of-4 S: scalar tail
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            if (qtok_base + lg * 4 + r < S) {
+              dsT[o + r] = ds4[r];
+              pdT[o + r] = pd4[r];
+            }
+        }
       }
     }
     // dQ = dS K : A = dSrow (this wave's LDS), B = Kt
@@ -598,6 +607,235 @@ __global__ __launch_bounds__(256, 2) void attn_bwd1_kernel(
         const int c = ct * 16 + lm;
         bf16x8 bkt = *(const bf16x8*)lds_at(
             lds, swz(32768 + c * 256 + (ks * 32 + lg * 8) * 2, c, 15));
+        qacc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(asr, bkt, qacc[ct], 0, 0, 0);
+      }
+    }
+    ushort_t* dq = dqkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qtok_base + lg * 4 + r;
+        const int c = ct * 16 + lm;
+        if (row < S) dq[(size_t)row * ts + c] = f32_to_bf16(qacc[ct][r]);
+      }
+    }
+  }
+}
+
+// Split-q variant of B1: (S+63)/64 blocks per (b,h), each covering 64 query
+// rows (one 16-row tile per wave, no qi loop). Exists because the one-block
+// B1 is latency-bound at 2 waves/SIMD — its 80 KB LDS caps residency at
+// 2 blocks/CU and its 201 VGPRs + 144 B/lane spill cap it at 2 waves/SIMD.
+// Here K^T is rebuilt IN PLACE over the dead V region after the score/dPd
+// MFMAs (one extra LDS round trip of K) and the per-wave dS buffer drops to
+// 16 rows, so the whole block fits 48 KB LDS with a smaller register
+// footprint -> 3 blocks/CU at 3+ waves/SIMD and 2x the workgroups.
+template <bool HAS_MASK>
+__global__ __launch_bounds__(256, 3) void attn_bwd1s_kernel(
+    const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ dout,
+    const ushort_t* __restrict__ mask, const float* __restrict__ m_io,
+    const float* __restrict__ l_io, ushort_t* __restrict__ pdT,
+    ushort_t* __restrict__ dsT, ushort_t* __restrict__ dqkv, int B, int S,
+    int h, float scale, float keep, uint64_t salt,
+    const unsigned long long* __restrict__ state) {
+  // LDS: V [128][64] sw7 @0 (16K, becomes Kt [64][128] sw15 after phase A),
+  // K [128][64] sw7 @16K, per-wave dS [16][128] sw15 @32K + w*4K. Total 48K.
+  extern __shared__ __attribute__((aligned(16))) char lds[];
+  const int tid = threadIdx.x;
+  const int l = tid & 63;
+  const int w = tid >> 6;
+  const int lm = l & 15;
+  const int lg = l >> 4;
+  const int QB = (S + 63) / 64;
+  const int bh = blockIdx.x / QB;
+  const int qb = blockIdx.x % QB;
+  const int b = bh / h;
+  const int hh = bh % h;
+  const int ts = 3 * h * ATT_D;
+  const int NT = (S + 15) / 16;
+  const uint64_t seed = salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull;
+  const float inv_keep = 1.f / keep;
+  const unsigned keep16 = keep_to_16(keep);
+
+  const ushort_t* qbase = qkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
+  const ushort_t* kbase = qbase + (size_t)h * ATT_D;
+  const ushort_t* vbase = qbase + (size_t)2 * h * ATT_D;
+  const ushort_t* dobase = dout + ((size_t)b * S * h + hh) * ATT_D;
+  const int dots = h * ATT_D;
+
+  const int qtok_base = qb * 64 + w * 16;
+  const bool active = qtok_base < S;
+
+  if (S < ATT_SMAX) {
+    for (int u = tid; u < (48 * 1024) / 16; u += 256)
+      *(ushort8_t*)lds_at(lds, u * 16) = (ushort8_t)(ushort_t)0;
+    __syncthreads();
+  }
+  // stage V [tok][64] sw7 @0 and K [tok][64] sw7 @16K
+  for (int u = tid; u < S * 8; u += 256) {
+    const int tok = u >> 3;
+    const int c16 = u & 7;
+    *(ushort8_t*)lds_at(lds, swz(0 + tok * 128 + c16 * 16, tok, 7)) =
+        *(const ushort8_t*)(vbase + (size_t)tok * ts + c16 * 8);
+    *(ushort8_t*)lds_at(lds, swz(16384 + tok * 128 + c16 * 16, tok, 7)) =
+        *(const ushort8_t*)(kbase + (size_t)tok * ts + c16 * 8);
+  }
+  // preload this wave's Q / dO fragments + mask column values
+  float mval[8];
+#pragma unroll
+  for (int kt = 0; kt < 8; ++kt) {
+    const int col = kt * 16 + lm;
+    float mv = 0.f;
+    if (HAS_MASK && col < S) mv = bf16_to_f32(mask[(size_t)b * S + col]);
+    mval[kt] = (col < S) ? mv : -3.0e38f;
+  }
+  int qtok = qtok_base + lm;
+  if (qtok >= S) qtok = S - 1;
+  bf16x8 aq[2], ado[2];
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    aq[ks] = *(const bf16x8*)(qbase + (size_t)qtok * ts + ks * 32 + lg * 8);
+    ado[ks] = *(const bf16x8*)(dobase + (size_t)qtok * dots + ks * 32 + lg * 8);
+  }
+  __syncthreads();
+
+  // phase A: scores + dPd
+  f32x4 sacc[8], dacc[8];
+#pragma unroll
+  for (int kt = 0; kt < 8; ++kt) {
+    sacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
+    dacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
+  }
+  if (active) {
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      if (kt >= NT) continue;
+      const int ktok = kt * 16 + lm;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 bk = *(const bf16x8*)lds_at(
+            lds, swz(16384 + ktok * 128 + (ks * 32 + lg * 8) * 2, ktok, 7));
+        sacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, sacc[kt], 0, 0, 0);
+        bf16x8 bv = *(const bf16x8*)lds_at(
+            lds, swz(0 + ktok * 128 + (ks * 32 + lg * 8) * 2, ktok, 7));
+        dacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[ks], bv, dacc[kt], 0, 0, 0);
+      }
+    }
+  }
+  __syncthreads();  // V dead from here
+  // rebuild Kt [c][tok] sw15 over the V region (K beyond S is zeroed above)
+  for (int u = tid; u < 128 * 8; u += 256) {
+    const int tok = u >> 3;
+    const int c16 = u & 7;
+    ushort8_t kv = *(const ushort8_t*)lds_at(
+        lds, swz(16384 + tok * 128 + c16 * 16, tok, 7));
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c16 * 8 + j;
+      *(ushort_t*)lds_at(lds, swz(0 + c * 256 + tok * 2, c, 15)) = kv[j];
+    }
+  }
+
+  if (active) {
+    // P, dP, rowdot, dS (p/pd recomputed in the store loop from live
+    // sacc/dacc + cached dropout hashes — same register discipline as B1)
+    float mrow[4], lrow[4], dot[4];
+    uint64_t zs[4][2];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qtok_base + lg * 4 + r;
+      const int rr = row < S ? row : S - 1;
+      mrow[r] = m_io[(size_t)bh * S + rr];
+      lrow[r] = 1.f / l_io[(size_t)bh * S + rr];
+      dot[r] = 0.f;
+      zs[r][0] = zs[r][1] = 0;
+      if (keep < 1.f) {
+        const uint64_t base = ((uint64_t)bh * S + row) * 16 + lm;
+        zs[r][0] = rng_hash(seed, base * 2);
+        zs[r][1] = rng_hash(seed, base * 2 + 1);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qtok_base + lg * 4 + r;
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        if (kt >= NT) continue;
+        const int col = kt * 16 + lm;
+        float p = __expf(sacc[kt][r] * scale + mval[kt] - mrow[r]) * lrow[r];
+        if (!(row < S && col < S)) p = 0.f;
+        float dp = dacc[kt][r];
+        if (keep < 1.f) {
+          const uint64_t zz = kt < 4 ? zs[r][0] : zs[r][1];
+          bool kbit = (unsigned)((zz >> (16 * (kt & 3))) & 0xFFFFu) < keep16;
+          dp = kbit ? dp * inv_keep : 0.f;
+        }
+        dot[r] += dp * p;
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) dot[r] += __shfl_xor(dot[r], off, 64);
+    }
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      if (kt >= NT) continue;
+      const int col = kt * 16 + lm;
+      ushort4_t ds4, pd4;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qtok_base + lg * 4 + r;
+        float p = __expf(sacc[kt][r] * scale + mval[kt] - mrow[r]) * lrow[r];
+        if (!(row < S && col < S)) p = 0.f;
+        float pd = p, dp = dacc[kt][r];
+        if (keep < 1.f) {
+          const uint64_t zz = kt < 4 ? zs[r][0] : zs[r][1];
+          bool kbit = (unsigned)((zz >> (16 * (kt & 3))) & 0xFFFFu) < keep16;
+          pd = kbit ? p * inv_keep : 0.f;
+          dp = kbit ? dp * inv_keep : 0.f;
+        }
+        const float ds = scale * p * (dp - dot[r]);
+        const int rl = lg * 4 + r;
+        *(ushort_t*)lds_at(lds, swz(32768 + w * 4096 + rl * 256 + col * 2, rl, 15)) =
+            f32_to_bf16(ds);
+        ds4[r] = f32_to_bf16(ds);
+        pd4[r] = f32_to_bf16(pd);
+      }
+      if (col < S) {
+        const size_t o = ((size_t)bh * S + col) * S + qtok_base + lg * 4;
+        if (qtok_base + lg * 4 + 3 < S) {
+          *(ushort4_t*)(dsT + o) = ds4;
+          *(ushort4_t*)(pdT + o) = pd4;
+        } else {
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            if (qtok_base + lg * 4 + r < S) {
+              dsT[o + r] = ds4[r];
+              pdT[o + r] = pd4[r];
+            }
+        }
+      }
+    }
+  }
+  __syncthreads();  // Kt complete before the dQ MFMAs read it
+
+  if (active) {
+    // dQ = dS K : A = this wave's dS rows (LDS), B = Kt
+    f32x4 qacc[4];
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) qacc[ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      if (ks * 32 >= S) break;
+      bf16x8 asr = *(const bf16x8*)lds_at(
+          lds, swz(32768 + w * 4096 + lm * 256 + (ks * 32 + lg * 8) * 2, lm, 15));
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        const int c = ct * 16 + lm;
+        bf16x8 bkt = *(const bf16x8*)lds_at(
+            lds, swz(0 + c * 256 + (ks * 32 + lg * 8) * 2, c, 15));
         qacc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(asr, bkt, qacc[ct], 0, 0, 0);
       }
     }
@@ -708,26 +946,49 @@ SKY_EXPORT int sky_attn_bwd(uint64_t stream, uint64_t qkv, uint64_t dout,
                             uint64_t state) {
   if (d != ATT_D || S > ATT_SMAX) return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
-  dim3 grid((unsigned)(B * h));
   bool hm = mask != 0;
-  if (hm)
-    hipLaunchKernelGGL((attn_bwd1_kernel<true>), grid, dim3(256), 80 * 1024, s,
-                       (const ushort_t*)qkv, (const ushort_t*)dout,
-                       (const ushort_t*)mask, (const float*)m,
-                       (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
-                       (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
-                       salt, (const unsigned long long*)state);
-  else
-    hipLaunchKernelGGL((attn_bwd1_kernel<false>), grid, dim3(256), 80 * 1024, s,
-                       (const ushort_t*)qkv, (const ushort_t*)dout,
-                       (const ushort_t*)mask, (const float*)m,
-                       (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
-                       (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
-                       salt, (const unsigned long long*)state);
-  hipLaunchKernelGGL(attn_bwd2_kernel, grid, dim3(256), 32 * 1024, s,
-                     (const ushort_t*)qkv, (const ushort_t*)dout,
-                     (const ushort_t*)pdT, (const ushort_t*)dsT,
-                     (ushort_t*)dqkv, (int)B, (int)S, (int)h);
+  // B1 variant: split-q (2 blocks per (b,h), 48K LDS, 3 waves/SIMD) is the
+  // measured default; SKY_ATTN_BWD1=wide selects the one-block variant.
+  const char* e = getenv("SKY_ATTN_BWD1");
+  const bool wide = e && e[0] == 'w';
+  if (wide) {
+    dim3 grid((unsigned)(B * h));
+    if (hm)
+      hipLaunchKernelGGL((attn_bwd1_kernel<true>), grid, dim3(256), 80 * 1024, s,
+                         (const ushort_t*)qkv, (const ushort_t*)dout,
+                         (const ushort_t*)mask, (const float*)m,
+                         (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
+                         (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
+                         salt, (const unsigned long long*)state);
+    else
+      hipLaunchKernelGGL((attn_bwd1_kernel<false>), grid, dim3(256), 80 * 1024, s,
+                         (const ushort_t*)qkv, (const ushort_t*)dout,
+                         (const ushort_t*)mask, (const float*)m,
+                         (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
+                         (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
+                         salt, (const unsigned long long*)state);
+  } else {
+    dim3 grid((unsigned)(B * h * ((S + 63) / 64)));
+    if (hm)
+      hipLaunchKernelGGL((attn_bwd1s_kernel<true>), grid, dim3(256), 48 * 1024, s,
+                         (const ushort_t*)qkv, (const ushort_t*)dout,
+                         (const ushort_t*)mask, (const float*)m,
+                         (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
+                         (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
+                         salt, (const unsigned long long*)state);
+    else
+      hipLaunchKernelGGL((attn_bwd1s_kernel<false>), grid, dim3(256), 48 * 1024, s,
+                         (const ushort_t*)qkv, (const ushort_t*)dout,
+                         (const ushort_t*)mask, (const float*)m,
+                         (const float*)lsum, (ushort_t*)pdT, (ushort_t*)dsT,
+                         (ushort_t*)dqkv, (int)B, (int)S, (int)h, scale, keep,
+                         salt, (const unsigned long long*)state);
+  }
+  hipLaunchKernelGGL(attn_bwd2_kernel, dim3((unsigned)(B * h)), dim3(256),
+                     32 * 1024, s, (const ushort_t*)qkv,
+                     (const ushort_t*)dout, (const ushort_t*)pdT,
+                     (const ushort_t*)dsT, (ushort_t*)dqkv, (int)B, (int)S,
+                     (int)h);
   LAUNCH_CHECK();
   return 0;
 }

@@ -208,7 +208,7 @@ def test_mfma_layout():
     assert torch.allclose(D, ref, atol=5e-2, rtol=5e-2), (D - ref).abs().max()
 
 
-@pytest.mark.parametrize("S", [128, 64, 48])
+@pytest.mark.parametrize("S", [128, 100, 64, 48])
 @pytest.mark.parametrize("with_mask", [True, False])
 def test_fused_attention_fwd_bwd(S, with_mask):
     torch.manual_seed(8)
@@ -574,11 +574,26 @@ def test_hblt_linear_bias_backward(dtype):
     )
 
 
-@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_hblt_linear_gelu_bf16_unsupported_fails_loudly():
+    """This hipBLASLt build rejects the GELU_AUX_BIAS epilogue for bf16 D
+    (HIPBLAS_STATUS_NOT_SUPPORTED, probed by tools/hblt_probe.py) — the op
+    must fail loudly, never fall back silently; the dispatch layer
+    (ops.linear_act) therefore gates the fusion to fp32."""
+    from skycomputing_amd.ops.functions import LinearGeluFn
+
+    x = torch.randn(64, 128, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(256, 128, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(256, dtype=torch.bfloat16, device="cuda")
+    with pytest.raises(RuntimeError, match="sky_hblt_linear_gelu_aux"):
+        LinearGeluFn.apply(x, w, b)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32])
 def test_hblt_linear_gelu(dtype):
-    """LinearGeluFn (GELU_AUX_BIAS epilogue). hipBLASLt's GELU is the tanh
-    approximation (max |erf-tanh| difference ~1.5e-3), so the forward is
-    compared at a tolerance above that; grads use the erf derivative."""
+    """LinearGeluFn (GELU_AUX_BIAS epilogue, fp32-only on this hipBLASLt
+    build). hipBLASLt's GELU is the tanh approximation (max |erf-tanh|
+    difference ~1.5e-3), so the forward is compared at a tolerance above
+    that; grads use the erf derivative."""
     from skycomputing_amd.ops.functions import LinearGeluFn
 
     torch.manual_seed(31)

@@ -73,3 +73,36 @@ def run_bwd(keep=0.9):
 
 print(f"attn_bwd  keep=0.9: {bench(lambda: run_bwd(0.9)):8.1f} us")
 print(f"attn_bwd  keep=1.0: {bench(lambda: run_bwd(1.0)):8.1f} us")
+
+# ---------------- backward A/B: wide (one block per bh) vs split-q ----------------
+import os
+
+dout = torch.randn_like(out)
+pdT = torch.empty(B, h, S, S, dtype=torch.bfloat16, device="cuda")
+dsT = torch.empty_like(pdT)
+dqkv = torch.empty_like(qkv)
+
+
+def run_bwd(keep=0.9):
+    check(lib.sky_attn_bwd(stream, ptr(qkv), ptr(dout), ptr(mask), ptr(m),
+                           ptr(l), ptr(pdT), ptr(dsT), ptr(dqkv),
+                           B, S, h, d, scale, keep, 123, 0), "bwd")
+
+
+results = {}
+for mode in ("wide", "split"):
+    if mode == "wide":
+        os.environ["SKY_ATTN_BWD1"] = "wide"
+    else:
+        os.environ.pop("SKY_ATTN_BWD1", None)
+    run_bwd()
+    torch.cuda.synchronize()
+    results[mode] = (dqkv.clone(), dsT.clone(), pdT.clone())
+    print(f"attn_bwd [{mode}] keep=0.9: {bench(lambda: run_bwd(0.9)):8.1f} us "
+          f"(bwd1+bwd2)")
+    print(f"attn_bwd [{mode}] keep=1.0: {bench(lambda: run_bwd(1.0)):8.1f} us")
+
+for name, a, b_ in zip(("dqkv", "dsT", "pdT"),
+                       results["wide"], results["split"]):
+    md = (a.float() - b_.float()).abs().max().item()
+    print(f"wide-vs-split max|d{''}| {name}: {md:.2e}")
