@@ -364,6 +364,10 @@ static int attn_launch(uint64_t stream, uint64_t qkv, uint64_t mask,
   dim3 grid((unsigned)(B * h));
   size_t lds_bytes = ATT_LDS_BYTES;
   bool hm = mask != 0;
+  // (a split-q forward variant — 2 blocks per (b,h), 32K LDS — measured
+  // SLOWER than this one-block kernel: 29.2 vs 27.2 us; the duplicated K
+  // staging doesn't pay without an occupancy gain, and the fwd register
+  // budget pins 2 waves/SIMD either way. See profiles/r01_notes.md.)
 #define ATT(HM, SML, PM)                                                      \
   hipLaunchKernelGGL((attn_fwd_kernel<HM, SML, PM>), grid, dim3(256),         \
                      lds_bytes, s, (const ushort_t*)qkv,                      \

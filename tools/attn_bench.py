@@ -106,3 +106,19 @@ for name, a, b_ in zip(("dqkv", "dsT", "pdT"),
                        results["wide"], results["split"]):
     md = (a.float() - b_.float()).abs().max().item()
     print(f"wide-vs-split max|d{''}| {name}: {md:.2e}")
+
+# ---------------- forward A/B: wide vs split-q ----------------
+fwd_res = {}
+for mode in ("wide", "split"):
+    if mode == "wide":
+        os.environ["SKY_ATTN_FWD"] = "wide"
+    else:
+        os.environ.pop("SKY_ATTN_FWD", None)
+    run_fwd()
+    torch.cuda.synchronize()
+    fwd_res[mode] = (out.clone(), m.clone(), l.clone())
+    print(f"attn_fwd [{mode}] keep=0.9: {bench(lambda: run_fwd(0.9)):8.1f} us")
+    print(f"attn_fwd [{mode}] keep=1.0: {bench(lambda: run_fwd(1.0)):8.1f} us")
+    print(f"attn_probs [{mode}] k=.9 : {bench(lambda: run_probs(0.9)):8.1f} us")
+for name, a, b_ in zip(("out", "m", "l"), fwd_res["wide"], fwd_res["split"]):
+    print(f"fwd wide-vs-split max|d| {name}: {(a.float()-b_.float()).abs().max().item():.2e}")
