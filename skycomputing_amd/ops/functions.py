@@ -33,12 +33,12 @@ def _stream() -> int:
     return torch.cuda.current_stream().cuda_stream
 
 
-_CS_SLABS = 256
+_CS_SLABS = 1024  # must match CS_SLABS / LN_SLABS in the HIP sources
 
 
 def _red_scratch(cols: int, pairs: int, device) -> torch.Tensor:
     """fp32 scratch for the two-stage column reductions:
-    [128 slabs][pairs*cols]."""
+    [_CS_SLABS slabs][pairs*cols]."""
     return torch.empty(_CS_SLABS * pairs * cols, dtype=torch.float32, device=device)
 
 

@@ -104,7 +104,7 @@ __global__ __launch_bounds__(256) void bias_gelu_bwd_dx_kernel(
 // no zero-init: stage 1 writes per-slab partials [nslabs][cols] fp32 to a
 // scratch buffer (thread owns 8 consecutive columns, 16 B loads); stage 2
 // reduces the slab axis and WRITES the result.
-#define CS_SLABS 256
+#define CS_SLABS 1024
 
 template <int DT, int BLOCK>
 __global__ __launch_bounds__(BLOCK) void colsum_part_kernel(
@@ -126,17 +126,19 @@ __global__ __launch_bounds__(BLOCK) void colsum_part_kernel(
   for (int j = 0; j < 8; ++j) scratch[(slab * cols8 + c8) * 8 + j] = s[j];
 }
 
-// two-level: block covers 16 columns x 16 slab-groups (each thread sums
-// nslabs/16 slabs of one column), LDS tree, then one write per column.
+// two-level: a 1024-thread block covers 16 columns x 64 slab-groups (each
+// thread sums nslabs/64 slabs of one column), parallel LDS tree, one write
+// per column. 16 waves/block keeps the chip's XCDs busy even though cols
+// only yields cols/16 blocks.
 template <int BLOCK, int DTOUT>
-__global__ __launch_bounds__(BLOCK) void colsum_final_kernel(
+__global__ __launch_bounds__(1024) void colsum_final_kernel(
     const float* __restrict__ scratch, void* __restrict__ out, int64_t cols,
     int64_t nslabs) {
-  __shared__ float lds[16][17];
+  __shared__ float lds[64][17];
   const int c = threadIdx.x & 15;
-  const int g = threadIdx.x >> 4;
+  const int g = threadIdx.x >> 4;  // 0..63
   const int64_t col = (int64_t)blockIdx.x * 16 + c;
-  const int64_t per = (nslabs + 15) / 16;
+  const int64_t per = (nslabs + 63) / 64;
   float s = 0.f;
   if (col < cols) {
     const int64_t y1 = min(nslabs, (int64_t)(g + 1) * per);
@@ -145,12 +147,11 @@ __global__ __launch_bounds__(BLOCK) void colsum_final_kernel(
   }
   lds[g][c] = s;
   __syncthreads();
-  if (g == 0 && col < cols) {
-    float t = 0.f;
-#pragma unroll
-    for (int j = 0; j < 16; ++j) t += lds[j][c];
-    store_elem<DTOUT>(out, col, t);
+  for (int st = 32; st >= 1; st >>= 1) {
+    if (g < st) lds[g][c] += lds[g + st][c];
+    __syncthreads();
   }
+  if (g == 0 && col < cols) store_elem<DTOUT>(out, col, lds[0][c]);
 }
 
 template <int DT, int BLOCK>
@@ -176,17 +177,24 @@ static void launch_colsum(hipStream_t s, const void* src, void* out,
   if (cols % 8 == 0 && scratch != nullptr) {
     constexpr int BLOCK = 128;
     const int64_t cols8 = cols / 8;
-    const int64_t nslabs = rows < CS_SLABS ? rows : CS_SLABS;
+    // adapt the slab count to the column-block count: ~1024 workgroups
+    // total fills the chip without inflating scratch traffic on wide
+    // matrices (which already parallelize across columns)
+    const int64_t gx = (cols8 + BLOCK - 1) / BLOCK;
+    int64_t nslabs = 1024 / gx;
+    if (nslabs < 128) nslabs = 128;
+    if (nslabs > CS_SLABS) nslabs = CS_SLABS;
+    if (nslabs > rows) nslabs = rows;
     const int64_t slab = (rows + nslabs - 1) / nslabs;
-    dim3 grid((unsigned)((cols8 + BLOCK - 1) / BLOCK), (unsigned)nslabs);
+    dim3 grid((unsigned)gx, (unsigned)nslabs);
     hipLaunchKernelGGL((colsum_part_kernel<DT, BLOCK>), grid, dim3(BLOCK), 0,
                        s, src, scratch, rows, cols8, slab);
     dim3 g2((unsigned)((cols + 15) / 16));
     if (dtout == DT_BF16)
-      hipLaunchKernelGGL((colsum_final_kernel<256, DT_BF16>), g2, dim3(256), 0,
+      hipLaunchKernelGGL((colsum_final_kernel<256, DT_BF16>), g2, dim3(1024), 0,
                          s, scratch, out, cols, nslabs);
     else
-      hipLaunchKernelGGL((colsum_final_kernel<256, DT_F32>), g2, dim3(256), 0,
+      hipLaunchKernelGGL((colsum_final_kernel<256, DT_F32>), g2, dim3(1024), 0,
                          s, scratch, out, cols, nslabs);
     return;
   }

@@ -260,7 +260,7 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_vec_kernel(
 // column-parallel dw/db, two-stage, no atomics, outputs need no zero-init:
 // stage 1 writes per-slab partials to scratch [nslabs][2*cols] fp32
 // (dw partial at [y][c], db partial at [y][cols+c]); stage 2 reduces.
-#define LN_SLABS 256
+#define LN_SLABS 1024
 
 template <int DT, bool HAS_RES, bool DROP, int BLOCK>
 __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_part_kernel(
@@ -315,18 +315,18 @@ __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_part_kernel(
   }
 }
 
-// two-level: block covers 16 columns x 16 slab-groups, LDS tree, one
-// write per column (see colsum_final_kernel).
+// two-level: a 1024-thread block covers 16 columns x 64 slab-groups,
+// parallel LDS tree, one write per column (see colsum_final_kernel).
 template <int BLOCK, int DTOUT>
-__global__ __launch_bounds__(BLOCK) void ln_bwd_wb_final_kernel(
+__global__ __launch_bounds__(1024) void ln_bwd_wb_final_kernel(
     const float* __restrict__ scratch, void* __restrict__ dw,
     void* __restrict__ db, int64_t cols, int64_t nslabs) {
-  __shared__ float ldsw[16][17];
-  __shared__ float ldsb[16][17];
+  __shared__ float ldsw[64][17];
+  __shared__ float ldsb[64][17];
   const int c = threadIdx.x & 15;
-  const int g = threadIdx.x >> 4;
+  const int g = threadIdx.x >> 4;  // 0..63
   const int64_t col = (int64_t)blockIdx.x * 16 + c;
-  const int64_t per = (nslabs + 15) / 16;
+  const int64_t per = (nslabs + 63) / 64;
   float sw = 0.f, sb = 0.f;
   if (col < cols) {
     const int64_t y1 = min(nslabs, (int64_t)(g + 1) * per);
@@ -338,12 +338,13 @@ __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_final_kernel(
   ldsw[g][c] = sw;
   ldsb[g][c] = sb;
   __syncthreads();
+  for (int st = 32; st >= 1; st >>= 1) {
+    if (g < st) { ldsw[g][c] += ldsw[g + st][c]; ldsb[g][c] += ldsb[g + st][c]; }
+    __syncthreads();
+  }
   if (g == 0 && col < cols) {
-    float tw = 0.f, tb = 0.f;
-#pragma unroll
-    for (int j = 0; j < 16; ++j) { tw += ldsw[j][c]; tb += ldsb[j][c]; }
-    store_elem<DTOUT>(dw, col, tw);
-    store_elem<DTOUT>(db, col, tb);
+    store_elem<DTOUT>(dw, col, ldsw[0][c]);
+    store_elem<DTOUT>(db, col, ldsb[0][c]);
   }
 }
 
@@ -448,9 +449,15 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
   if (cols % 8 == 0 && scratch != 0) {
     constexpr int BLOCK = 128;
     const int64_t cols8 = cols / 8;
-    const int64_t nslabs = rows < LN_SLABS ? rows : LN_SLABS;
+    // ~1024 workgroups total (see launch_colsum): more slabs only where
+    // the column dimension is too narrow to fill the chip
+    const int64_t gx = (cols8 + BLOCK - 1) / BLOCK;
+    int64_t nslabs = 1024 / gx;
+    if (nslabs < 128) nslabs = 128;
+    if (nslabs > LN_SLABS) nslabs = LN_SLABS;
+    if (nslabs > rows) nslabs = rows;
     const int64_t slab = (rows + nslabs - 1) / nslabs;
-    dim3 grid((unsigned)((cols8 + BLOCK - 1) / BLOCK), (unsigned)nslabs);
+    dim3 grid((unsigned)gx, (unsigned)nslabs);
 #define LNWBP(DT, HR, DR)                                                      \
   hipLaunchKernelGGL((ln_bwd_wb_part_kernel<DT, HR, DR, BLOCK>), grid,         \
                      dim3(BLOCK), 0, s, (const void*)dy, (const void*)x,       \
@@ -465,11 +472,11 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
 #undef LNWBP
     dim3 g2((unsigned)((cols + 15) / 16));
     if (dt == DT_BF16)
-      hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256, DT_BF16>), g2, dim3(256),
+      hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256, DT_BF16>), g2, dim3(1024),
                          0, s, (const float*)scratch, (void*)dw, (void*)db,
                          cols, nslabs);
     else
-      hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256, DT_F32>), g2, dim3(256),
+      hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256, DT_F32>), g2, dim3(1024),
                          0, s, (const float*)scratch, (void*)dw, (void*)db,
                          cols, nslabs);
   } else {
