@@ -38,3 +38,30 @@ def test_bench_json_contract(tmp_path):
     for key in ("model", "global_batch", "seq_len", "parallelism"):
         assert key in cfg, key
     assert json.loads(out.read_text()) == d
+
+
+def test_bench_8rank_driver_invocation(tmp_path):
+    """Exactly the command shape the driver uses for the 8-GPU scaling run
+    (torch.distributed.run, one rank per device) — on gloo/CPU with a tiny
+    model, so the full multi-rank bench path (partition, pipeline schedule,
+    microbatching, JSON emission) is exercised before it ever meets 8 GPUs."""
+    out = tmp_path / "b8.json"
+    env = dict(os.environ)
+    port = str(21500 + os.getpid() % 20000)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", port,
+         "bench.py", "--gpus", "8", "--layers", "4", "--batch", "8",
+         "--seq", "16", "--steps", "2", "--warmup", "1",
+         "--json-out", str(out)],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=560,
+    )
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    json_lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1, r.stdout[-2000:]
+    d = json.loads(json_lines[0])
+    assert d["n_gpus"] == 8
+    assert d["config"]["parallelism"] == "pp8"
+    assert d["config"]["microbatches"] == 8
+    assert d["value"] > 0
