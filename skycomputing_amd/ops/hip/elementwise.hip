@@ -100,6 +100,39 @@ __global__ __launch_bounds__(256) void bias_gelu_bwd_dx_kernel(
   }
 }
 
+// fused dx + db column partials: computes dx = dy * gelu'(pre) AND
+// accumulates its per-slab column sums in the same pass, so db needs no
+// re-read of the 32 MB dx tensor (stage 2 = colsum_final over scratch).
+// Same (column-block x row-slab) grid as colsum_part_kernel.
+template <int DT, bool HAS_B, int BLOCK>
+__global__ __launch_bounds__(BLOCK) void bias_gelu_bwd_part_kernel(
+    const void* __restrict__ dy, const void* __restrict__ x,
+    const void* __restrict__ b, void* __restrict__ dx,
+    float* __restrict__ scratch, int64_t rows, int64_t cols8,
+    int64_t rows_per_slab) {
+  const int64_t c8 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  if (c8 >= cols8) return;
+  const int64_t slab = blockIdx.y;
+  const int64_t r0 = slab * rows_per_slab;
+  const int64_t r1 = min(rows, r0 + rows_per_slab);
+  float bv[8];
+  if (HAS_B) Vec8<DT>::load(b, c8, bv);
+  float s[8] = {0.f};
+  for (int64_t r = r0; r < r1; ++r) {
+    float v[8], d[8];
+    Vec8<DT>::load(x, r * cols8 + c8, v);
+    Vec8<DT>::load(dy, r * cols8 + c8, d);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      d[j] *= gelu_grad_f(HAS_B ? v[j] + bv[j] : v[j]);
+      s[j] += d[j];
+    }
+    Vec8<DT>::store(dx, r * cols8 + c8, d);
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) scratch[(slab * cols8 + c8) * 8 + j] = s[j];
+}
+
 // db[c] = sum over rows of dx[r, c] — two-stage, no atomics, outputs need
 // no zero-init: stage 1 writes per-slab partials [nslabs][cols] fp32 to a
 // scratch buffer (thread owns 8 consecutive columns, 16 B loads); stage 2
@@ -228,6 +261,38 @@ SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
   hipStream_t s = (hipStream_t)stream;
   int64_t n = rows * cols;
   const bool hb = b != 0;
+  if (cols % 8 == 0 && scratch != 0) {
+    // fused one-pass dx + db partials (see bias_gelu_bwd_part_kernel)
+    constexpr int BLOCK = 128;
+    const int64_t cols8 = cols / 8;
+    const int64_t gx = (cols8 + BLOCK - 1) / BLOCK;
+    int64_t nslabs = 1024 / gx;
+    if (nslabs < 128) nslabs = 128;
+    if (nslabs > CS_SLABS) nslabs = CS_SLABS;
+    if (nslabs > rows) nslabs = rows;
+    const int64_t slab = (rows + nslabs - 1) / nslabs;
+    dim3 grid((unsigned)gx, (unsigned)nslabs);
+    auto launch_part = [&](auto kern) {
+      hipLaunchKernelGGL(kern, grid, dim3(BLOCK), 0, s, (const void*)dy,
+                         (const void*)x, (const void*)b, (void*)dx,
+                         (float*)scratch, rows, cols8, slab);
+    };
+    if (dt == DT_F32)
+      hb ? launch_part(bias_gelu_bwd_part_kernel<DT_F32, true, BLOCK>)
+         : launch_part(bias_gelu_bwd_part_kernel<DT_F32, false, BLOCK>);
+    else
+      hb ? launch_part(bias_gelu_bwd_part_kernel<DT_BF16, true, BLOCK>)
+         : launch_part(bias_gelu_bwd_part_kernel<DT_BF16, false, BLOCK>);
+    dim3 g2((unsigned)((cols + 15) / 16));
+    if (dt == DT_BF16)
+      hipLaunchKernelGGL((colsum_final_kernel<256, DT_BF16>), g2, dim3(1024),
+                         0, s, (float*)scratch, (void*)db, cols, nslabs);
+    else
+      hipLaunchKernelGGL((colsum_final_kernel<256, DT_F32>), g2, dim3(1024),
+                         0, s, (float*)scratch, (void*)db, cols, nslabs);
+    LAUNCH_CHECK();
+    return 0;
+  }
   if (cols % 8 == 0) {
     int64_t n8 = n / 8;
     unsigned grid = (unsigned)((n8 + 255) / 256);
