@@ -134,9 +134,13 @@ class GlueDataset(Dataset):
         tok = BertTokenizerFast(vocab_file=vocab, do_lower_case=True)
         examples = self._read_examples(data_dir, task, split)
         label_map = {l: i for i, l in enumerate(self.TASK_LABELS[task])}
+        texts = [a for a, _, _ in examples]
+        pairs = [b for _, b, _ in examples]
+        # single-sentence tasks (sst-2, cola) must pass text_pair=None, not
+        # a list of empty strings
+        pair_arg = pairs if any(pairs) else None
         enc = tok(
-            [a for a, _, _ in examples],
-            [b if b else None for _, b, _ in examples],
+            texts, pair_arg,
             padding="max_length", truncation=True, max_length=max_seq_length,
             return_tensors="pt",
         )

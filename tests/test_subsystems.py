@@ -106,6 +106,24 @@ def test_glue_dataset_tokenizes_from_disk(tmp_path):
     assert mask.sum() >= 7
 
 
+def test_glue_dataset_single_sentence_tasks(tmp_path):
+    """sst-2/cola have no second sentence — tokenization must pass
+    text_pair=None, not a list of empty strings (regression test)."""
+    vocab = ["[PAD]", "[UNK]", "[CLS]", "[SEP]", "[MASK]", "the", "cat", "sat"]
+    (tmp_path / "vocab.txt").write_text("\n".join(vocab) + "\n")
+    (tmp_path / "train.tsv").write_text(
+        "sentence\tlabel\nthe cat sat\t0\nthe cat\t1\n")
+    try:
+        ds = GlueDataset(str(tmp_path), task="sst-2", max_seq_length=12)
+    except ImportError:
+        pytest.skip("transformers unavailable")
+    assert len(ds) == 2
+    (ids, mask, tids), label = ds[1]
+    assert ids.shape == (12,)
+    assert int(label) == 1
+    assert int(tids.max()) == 0  # single segment
+
+
 def test_random_token_generator():
     g = RandomTokenGenerator(batch_size=2, seq_len=4, vocab_size=10)
     ids, mask, tids = g.generate()
