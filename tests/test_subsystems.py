@@ -169,3 +169,56 @@ def test_config_registry_hooks_buildable():
     assert isinstance(
         build_hook(dict(layer_type="CheckpointHook", save_path="/tmp/x")), CheckpointHook
     )
+
+
+def test_dataloader_generator_advances_and_wraps():
+    """The reference's DataloaderGenerator always returned the FIRST batch
+    (scaelum/dataset/data_generator.py:30-34); ours must advance through
+    the loader and wrap around at exhaustion."""
+    from skycomputing_amd.dataset.data_generator import DataloaderGenerator
+
+    gen = DataloaderGenerator(dict(
+        batch_size=4,
+        dataset_cfg=dict(type="RandomMlpDataset", size=8, dim=6, num_class=3),
+    ))
+    b1 = gen.generate()
+    b2 = gen.generate()
+    assert b1.shape == (4, 6)
+    assert not torch.allclose(b1, b2)  # advanced, not the first batch again
+    b3 = gen.generate()  # loader exhausted (8/4 = 2 batches) -> wraps
+    assert torch.allclose(b1, b3)
+
+
+def test_estimator_benchmark_speed_cpu():
+    from skycomputing_amd.dataset.data_generator import RandomTensorGenerator
+    from skycomputing_amd.dynamics.estimator import Estimator
+
+    model = torch.nn.Sequential(torch.nn.Linear(16, 16), torch.nn.ReLU())
+    gen = RandomTensorGenerator(size=(4, 16))
+    t = Estimator.benchmark_speed(model, gen, iterations=3, warmup=1,
+                                  backward=True, device=torch.device("cpu"))
+    assert t > 0
+
+
+def test_logger_writes_per_rank_file(tmp_path):
+    from skycomputing_amd.logger import Logger
+
+    log = tmp_path / "run.log"
+    lg = Logger(log_file=str(log), rank=3)
+    lg.info("hello world")
+    text = log.read_text()
+    assert "hello world" in text
+
+
+def test_hook_cadence_helpers():
+    from skycomputing_amd.runner.hooks import Hook
+
+    class R:  # 0-based counters: epoch 3 = the 4th epoch, iter 9 = the 10th
+        epoch = 3
+        iter = 9
+
+    assert Hook.every_n_epochs(R, 2)
+    assert not Hook.every_n_epochs(R, 3)
+    assert Hook.every_n_iters(R, 5)
+    assert not Hook.every_n_iters(R, 4)
+    assert not Hook.every_n_iters(R, 0)  # disabled cadence never fires
