@@ -262,11 +262,15 @@ SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
   int64_t n = rows * cols;
   const bool hb = b != 0;
   if (cols % 8 == 0 && scratch != 0) {
-    // fused one-pass dx + db partials (see bias_gelu_bwd_part_kernel)
+    // fused one-pass dx + db partials (see bias_gelu_bwd_part_kernel).
+    // This kernel moves 3 tensors (dy, x read; dx write), so unlike the
+    // read-only colsum part it wants FULL thread count (~4096 waves, short
+    // row chains); the extra scratch volume is small against 96 MB of
+    // main traffic.
     constexpr int BLOCK = 128;
     const int64_t cols8 = cols / 8;
     const int64_t gx = (cols8 + BLOCK - 1) / BLOCK;
-    int64_t nslabs = 1024 / gx;
+    int64_t nslabs = 4096 / gx;
     if (nslabs < 128) nslabs = 128;
     if (nslabs > CS_SLABS) nslabs = CS_SLABS;
     if (nslabs > rows) nslabs = rows;
