@@ -35,6 +35,13 @@ DEV void glds16(const void* g, void* l) {
   __builtin_amdgcn_global_load_lds((gas_u32)g, (las_u32)l, 16, 0, 0);
 }
 
+// XOR swizzle for the LDS tile images (rows are 128 B = 8 x 16 B units):
+// without it every lane of a fragment-read quad starts on the SAME bank
+// (row stride 128 B), serializing ds_read_b128 16-way. The glds path
+// applies the swizzle on the SOURCE column instead (lane-linear LDS image
+// == swizzled layout, guide rule 21).
+DEV int gswz(int byte_addr, int row) { return byte_addr ^ ((row & 7) << 4); }
+
 // epilogue modes
 enum { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_GELU = 2 };
 
@@ -75,11 +82,16 @@ __global__ __launch_bounds__(GM_BLOCK) void gemm_kernel(
     for (int it = 0; it < 4; ++it) {
       const int u = it * 256 + tid;
       const int row = u >> 3, c16 = u & 7;
-      const ushort_t* g = src + (size_t)(row0 + (row < rows ? row : rows - 1)) * ld + k0 + c16 * 8;
-      if (USE_GLDS)
+      if (USE_GLDS) {
+        // source-side swizzle: fetch the column that belongs at this
+        // lane-linear LDS slot under the XOR layout
+        const int c16s = c16 ^ (row & 7);
+        const ushort_t* g = src + (size_t)(row0 + (row < rows ? row : rows - 1)) * ld + k0 + c16s * 8;
         glds16(g, dst + u * 16);
-      else
-        *(ushort8_t*)(dst + u * 16) = *(const ushort8_t*)g;
+      } else {
+        const ushort_t* g = src + (size_t)(row0 + (row < rows ? row : rows - 1)) * ld + k0 + c16 * 8;
+        *(ushort8_t*)(dst + gswz(row * 128 + c16 * 16, row)) = *(const ushort8_t*)g;
+      }
     }
   };
   auto stage_trans = [&](const ushort_t* src, int ld, int k0, int col0,
@@ -91,8 +103,10 @@ __global__ __launch_bounds__(GM_BLOCK) void gemm_kernel(
       const int r = u >> 4, c8 = u & 15;
       ushort8_t v = *(const ushort8_t*)(src + (size_t)(k0 + r) * ld + col0 + c8 * 8);
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *(ushort_t*)(dst + ((c8 * 8 + j) * GM_BK + r) * 2) = v[j];
+      for (int j = 0; j < 8; ++j) {
+        const int row = c8 * 8 + j;
+        *(ushort_t*)(dst + gswz(row * 128 + r * 2, row)) = v[j];
+      }
     }
   };
 
@@ -133,9 +147,9 @@ __global__ __launch_bounds__(GM_BLOCK) void gemm_kernel(
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int am = wr * 64 + i * 16 + lm;
-        af[i] = *(const bf16x8*)(abuf + (am * GM_BK + ks * 32 + lg * 8) * 2);
+        af[i] = *(const bf16x8*)(abuf + gswz(am * 128 + (ks * 32 + lg * 8) * 2, am));
         const int bn = wc * 64 + i * 16 + lm;
-        bf[i] = *(const bf16x8*)(bbuf + (bn * GM_BK + ks * 32 + lg * 8) * 2);
+        bf[i] = *(const bf16x8*)(bbuf + gswz(bn * 128 + (ks * 32 + lg * 8) * 2, bn));
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
