@@ -12,7 +12,7 @@ from skycomputing_amd.builder import (
 )
 from skycomputing_amd.config import Config, load_config
 from skycomputing_amd.dynamics import Estimator, ModelBenchmarker
-from skycomputing_amd.registry import LAYER, Registry
+from skycomputing_amd.registry import Registry
 
 from .helpers import tiny_bert_cfg
 

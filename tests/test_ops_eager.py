@@ -6,8 +6,6 @@ against these in fp32.
 
 from __future__ import annotations
 
-import math
-
 import torch
 
 from skycomputing_amd import ops

@@ -246,8 +246,7 @@ def test_fused_attention_dropout_consistency():
     """With dropout on, E[out] matches the no-dropout output and backward
     regenerates exactly the forward mask (zero positions align)."""
     torch.manual_seed(9)
-    from skycomputing_amd.ops import hiplib as hl
-    from skycomputing_amd.ops.functions import FusedAttentionFn, rng_state
+    from skycomputing_amd.ops.functions import FusedAttentionFn
     from skycomputing_amd.ops.hiplib import check, ptr
 
     B, S, h, d = 2, 128, 4, 64

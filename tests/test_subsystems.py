@@ -3,20 +3,17 @@ datasets, optimizer, parameter server."""
 
 from __future__ import annotations
 
-import os
-
 import pytest
 import torch
 
 from skycomputing_amd.dataset import (
-    GlueDataset, RandomMlpDataset, RandomTokenGenerator, SyntheticGlueDataset,
+    GlueDataset, RandomTokenGenerator, SyntheticGlueDataset,
 )
 from skycomputing_amd.dynamics import ParameterServer, Worker, WorkerManager
 from skycomputing_amd.optim import FusedSGD
 from skycomputing_amd.stimulator import Stimulator
 from skycomputing_amd.timer import DeviceTimer, DistributedTimer
 
-from .helpers import tiny_bert_cfg
 
 
 def test_worker_roundtrip_and_manager():

@@ -1,7 +1,7 @@
 import sys, torch
 sys.path.insert(0, "/root/repo")
 from skycomputing_amd.ops import hiplib
-from skycomputing_amd.ops.hiplib import check, ptr
+from skycomputing_amd.ops.hiplib import ptr
 lib = hiplib.require()
 src = torch.arange(2048, dtype=torch.float32).to(torch.bfloat16).cuda()
 stream = torch.cuda.current_stream().cuda_stream
