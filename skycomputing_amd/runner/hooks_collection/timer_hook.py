@@ -14,6 +14,12 @@ from ..hooks import Hook
 
 @HOOKS.register_module
 class TimerHook(Hook):
+    def __init__(self, trace_path: str | None = None):
+        """``trace_path``: optional chrome://tracing JSON written at run
+        end (per rank: ``{trace_path}.rank{r}.json``) from the recorded
+        iteration intervals."""
+        self.trace_path = trace_path
+
     def before_run(self, runner):
         runner.timer.clean()
 
@@ -38,4 +44,9 @@ class TimerHook(Hook):
             )
 
     def after_run(self, runner):
+        if self.trace_path:
+            rank = getattr(runner.comm, "rank", 0) if getattr(runner, "comm", None) else 0
+            runner.timer.export_chrome_trace(
+                f"{self.trace_path}.rank{rank}.json", rank=rank
+            )
         runner.timer.clean()

@@ -219,3 +219,18 @@ def test_hook_cadence_helpers():
     assert Hook.every_n_iters(R, 5)
     assert not Hook.every_n_iters(R, 4)
     assert not Hook.every_n_iters(R, 0)  # disabled cadence never fires
+
+
+def test_chrome_trace_export(tmp_path):
+    import json
+    import time
+
+    dt = DistributedTimer()
+    for _ in range(2):
+        dt.add_timestamp("iter"); time.sleep(0.002); dt.add_timestamp("iter")
+    path = str(tmp_path / "trace.json")
+    dt.export_chrome_trace(path, rank=3)
+    trace = json.loads(open(path).read())
+    ev = trace["traceEvents"]
+    assert len(ev) == 2
+    assert all(e["ph"] == "X" and e["tid"] == 3 and e["dur"] > 0 for e in ev)
