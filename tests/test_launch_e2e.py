@@ -251,3 +251,30 @@ def test_reallocation_mid_training(tmp_path):
     batch = (ids, _t.zeros_like(ids), _t.ones_like(ids))
     labels = _t.randint(0, 3, (8,), generator=g)
     run_multiprocess(_replan_worker, 2, 29910, layer_cfgs, batch, labels, str(tmp_path))
+
+
+def test_baseline_configs_load():
+    """The five BASELINE.json configurations ship as ready-to-run config
+    files; each must parse through the config loader with the expected
+    layer count and allocation mode."""
+    import glob
+
+    from skycomputing_amd.config import load_config
+
+    expect = {
+        "bert24_even_cpu.py": (24, "even", False),
+        "bert24_optimal_8gpu.py": (24, "optimal", False),
+        "bert160_dynamic_8gpu.py": (160, "dynamic", False),
+        "bert96_stimulate_8gpu.py": (96, "optimal", True),
+        "bert320_optimal_8gpu.py": (320, "optimal", False),
+    }
+    files = {os.path.basename(f): f
+             for f in glob.glob(os.path.join(REPO, "experiment/configs/*.py"))}
+    assert set(files) == set(expect)
+    for name, (layers, mode, stim) in expect.items():
+        cfg = load_config(files[name])
+        assert cfg.model_config["num_encoder_layers"] == layers
+        assert cfg.allocator_config["mode"] == mode
+        assert bool(cfg.allocator_config.get("stimulate")) == stim
+        assert cfg.data_config["batch_size"] == 32
+        assert cfg.train_config["max_iter"] == 30
