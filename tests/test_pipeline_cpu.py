@@ -211,3 +211,59 @@ def _val_worker(rank, world_size, layer_cfgs, out_dir):
 
 def test_runner_val_pipeline():
     run_multiprocess(_val_worker, 2, 29950, tiny_bert_cfg(1), ".")
+
+
+def _four_stage_worker(rank, world_size, layer_cfgs, batch, labels, lr, steps, M, out_dir):
+    torch.manual_seed(1234)
+    from skycomputing_amd.builder import build_module_from_cfg
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import PartitionPlan, PipelineEngine, init_distributed
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    full = build_module_from_cfg(layer_cfgs, record_forward_time=False)
+    L = len(layer_cfgs)
+    base, rem = divmod(L, world_size)
+    bounds = [0]
+    for i in range(world_size):
+        bounds.append(bounds[-1] + base + (1 if i < rem else 0))
+    plan = PartitionPlan(
+        stage_ranks=list(range(world_size)),
+        ranges=[(bounds[i], bounds[i + 1]) for i in range(world_size)],
+    )
+    engine = PipelineEngine(comm, layer_cfgs, plan, loss_fn=torch.nn.CrossEntropyLoss(),
+                            stage_kwargs=dict(record_forward_time=False))
+    start, end = plan.ranges[engine.stage_idx]
+    engine.stage.load_layer_state_dicts(
+        [
+            {k: v.detach().clone() for k, v in full.module[i].state_dict().items()}
+            for i in range(start, end)
+        ]
+    )
+    opt = FusedSGD(engine.parameters(), lr=lr)
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = engine.run_iteration(batch, labels, num_microbatches=M, schedule="gpipe")
+        opt.step()
+        losses.append(loss)
+    if rank == 0:
+        np.save(f"{out_dir}/losses4.npy", np.array(losses, dtype=np.float64))
+    comm.barrier()
+    from skycomputing_amd.parallel import destroy
+
+    destroy()
+
+
+def test_four_stage_gpipe_m8_matches_local(tmp_path):
+    """The driver's SCALE shape: even split over 4 stages, gpipe, M=8 —
+    per-step losses must match the single-process reference exactly."""
+    layer_cfgs = tiny_bert_cfg(2)
+    batch, labels = _make_batch()
+    lr, steps, M = 0.05, 3, 8
+    ref = _single_process_reference(layer_cfgs, batch, labels, lr, steps, M)
+    run_multiprocess(
+        _four_stage_worker, 4, 29720, layer_cfgs, batch, labels, lr, steps, M, str(tmp_path)
+    )
+    got = np.load(f"{tmp_path}/losses4.npy")
+    assert np.allclose(got, np.array(ref), rtol=1e-4, atol=1e-5), (got, ref)
+    assert got[-1] < got[0]
