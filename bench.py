@@ -35,6 +35,8 @@ def parse_args():
     p.add_argument("--allocate", default="even", choices=["even", "dynamic", "optimal"])
     p.add_argument("--microbatches", type=int, default=0)
     p.add_argument("--schedule", default="gpipe", choices=["gpipe", "sequential", "1f1b"])
+    p.add_argument("--virtual-stages", type=int, default=1,
+                   help="interleaved chunks per rank (N>1; eager, no graphs)")
     p.add_argument("--slowdowns", default="")
     p.add_argument("--stimulate", action="store_true")
     p.add_argument("--dropout", type=float, default=0.1)
@@ -103,14 +105,30 @@ def main():
         plan = PartitionPlan.from_dict(comm.broadcast_object(plan_d, src=0))
 
     sd = slowdowns[rank] if rank < len(slowdowns) else 0.0
-    engine = PipelineEngine(
-        comm, layer_cfgs, plan,
-        loss_fn=lambda logits, labels: torch.nn.functional.cross_entropy(
-            logits.float(), labels
-        ),
-        dtype=dtype,
-        stage_kwargs=dict(record_forward_time=bool(sd > 0), slowdown=sd),
-    )
+    virtual = args.virtual_stages if world > 1 else 1
+    if virtual > 1:
+        from skycomputing_amd.parallel.interleaved import (
+            InterleavedPipelineEngine, build_interleaved_plan,
+        )
+
+        plan = build_interleaved_plan(L, world, virtual)
+        engine = InterleavedPipelineEngine(
+            comm, layer_cfgs, plan,
+            loss_fn=lambda logits, labels: torch.nn.functional.cross_entropy(
+                logits.float(), labels
+            ),
+            dtype=dtype,
+            stage_kwargs=dict(record_forward_time=bool(sd > 0), slowdown=sd),
+        )
+    else:
+        engine = PipelineEngine(
+            comm, layer_cfgs, plan,
+            loss_fn=lambda logits, labels: torch.nn.functional.cross_entropy(
+                logits.float(), labels
+            ),
+            dtype=dtype,
+            stage_kwargs=dict(record_forward_time=bool(sd > 0), slowdown=sd),
+        )
     opt = FusedSGD(engine.parameters(), lr=1e-3)
 
     ds = SyntheticGlueDataset(size=args.batch * 4, max_seq_length=args.seq, seed=7)
@@ -118,7 +136,8 @@ def main():
     batches = list(loader)
 
     graphed = None
-    if use_cuda and sd == 0 and not args.no_graph and args.schedule == "gpipe":
+    if (use_cuda and sd == 0 and not args.no_graph
+            and args.schedule == "gpipe" and virtual == 1):
         data0, labels0 = batches[0]
         if world == 1 and M == 1:
             from skycomputing_amd.parallel.graph import GraphedTrainStep
@@ -147,7 +166,11 @@ def main():
             graphed.step(data, labels)
             return
         opt.zero_grad(set_to_none=True)
-        engine.run_iteration(data, labels, num_microbatches=M, schedule=args.schedule)
+        if virtual > 1:
+            engine.run_iteration(data, labels, num_microbatches=M)
+        else:
+            engine.run_iteration(data, labels, num_microbatches=M,
+                                 schedule=args.schedule)
         opt.step()
 
     for i in range(args.warmup):
@@ -183,7 +206,8 @@ def main():
                 "model": f"bert-{args.layers}L-H1024-A16",
                 "global_batch": args.batch,
                 "seq_len": args.seq,
-                "parallelism": f"pp{world}",
+                "parallelism": (f"pp{world}x{virtual}" if virtual > 1
+                                 else f"pp{world}"),
                 "microbatches": M,
                 "schedule": args.schedule,
                 "allocate": args.allocate,

@@ -28,6 +28,19 @@ import torch.distributed as dist
 _CTX: "CommContext | None" = None
 
 
+class _RetainingWork:
+    """Wraps a dist work handle, keeping the sent buffer alive until
+    wait()."""
+
+    def __init__(self, work, buf):
+        self._work = work
+        self._buf = buf
+
+    def wait(self):
+        self._work.wait()
+        self._buf = None
+
+
 class CommContext:
     def __init__(self, backend: str, device: torch.device, ctrl_group):
         self.backend = backend
@@ -105,6 +118,29 @@ class CommContext:
 
     def cached_recv_meta(self, src: int, key: str):
         return self._meta_cache_recv.get((src, key))
+
+    def isend_tensors(self, tensors: list, dst: int, key: str,
+                      blocking: bool = False):
+        """Non-blocking sends (the interleaved engine's transport): returns
+        work handles that also retain the sent buffers until wait(). The
+        channel meta must already be cached (do one ``blocking=True``
+        iteration first — it delegates to the handshaking send_tensors)."""
+        if blocking:
+            self.send_tensors(tensors, dst, key)
+            return []
+        metas = [
+            (tuple(t.shape), t.dtype, bool(t.requires_grad)) for t in tensors
+        ]
+        cached = self._meta_cache_send.get((dst, key))
+        assert cached == metas, (
+            f"channel ({dst},{key}) meta changed or missing; run a blocking "
+            "handshake iteration first"
+        )
+        works = []
+        for t in tensors:
+            buf = t.detach().contiguous()
+            works.append(_RetainingWork(dist.isend(buf, dst=dst), buf))
+        return works
 
     def recv_tensors_into(self, bufs: list, src: int):
         """Receive into preallocated (static) buffers — the graphed

@@ -65,3 +65,24 @@ def test_bench_8rank_driver_invocation(tmp_path):
     assert d["config"]["parallelism"] == "pp8"
     assert d["config"]["microbatches"] == 8
     assert d["value"] > 0
+
+
+def test_bench_virtual_stages_cpu(tmp_path):
+    """Opt-in interleaved virtual stages through the bench entrypoint
+    (4 ranks x 2 chunks on gloo/CPU)."""
+    out = tmp_path / "bv.json"
+    env = dict(os.environ)
+    port = str(22500 + os.getpid() % 20000)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", port,
+         "bench.py", "--gpus", "4", "--layers", "4", "--batch", "8",
+         "--seq", "16", "--steps", "2", "--warmup", "1",
+         "--virtual-stages", "2", "--json-out", str(out)],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=560,
+    )
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][0])
+    assert d["config"]["parallelism"] == "pp4x2"
+    assert d["value"] > 0
