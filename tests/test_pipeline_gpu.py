@@ -218,3 +218,53 @@ def test_graphed_pipeline_step_single_stage():
     losses = [g2.step(inputs, labels) for _ in range(10)]
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0] * 0.9, losses
+
+
+def test_interleaved_engine_single_rank_v2():
+    """InterleavedPipelineEngine at world=1, v=2 (both chunks local, the
+    mailbox/gradbox path) on bf16 HIP kernels: must match the one-chunk
+    eager pipeline's loss and train."""
+    torch.manual_seed(5)
+    from skycomputing_amd.models import bert_pipeline_config
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import PartitionPlan, PipelineEngine, init_distributed
+    from skycomputing_amd.parallel.interleaved import (
+        InterleavedPipelineEngine, build_interleaved_plan,
+    )
+
+    comm = init_distributed()
+    cfg = dict(hidden_size=256, num_attention_heads=4, intermediate_size=1024,
+               vocab_size=2000, hidden_dropout_prob=0.0,
+               attention_probs_dropout_prob=0.0)
+    cfgs = bert_pipeline_config(2, cfg)
+    ids = torch.randint(0, 2000, (16, 32))
+    inputs = (ids, torch.zeros_like(ids), torch.ones_like(ids))
+    labels = torch.randint(0, 3, (16,))
+    lf = lambda lg, lb: torch.nn.functional.cross_entropy(lg.float(), lb.to(lg.device))  # noqa: E731
+
+    torch.manual_seed(6)
+    ref_engine = PipelineEngine(
+        comm, cfgs, PartitionPlan(stage_ranks=[0], ranges=[(0, len(cfgs))]),
+        loss_fn=lf, dtype=torch.bfloat16,
+        stage_kwargs=dict(record_forward_time=False),
+    )
+    ref_loss = ref_engine.run_iteration(inputs, labels, num_microbatches=4,
+                                        schedule="gpipe")
+
+    torch.manual_seed(6)
+    plan = build_interleaved_plan(len(cfgs), 1, 2)
+    engine = InterleavedPipelineEngine(
+        comm, cfgs, plan, loss_fn=lf, dtype=torch.bfloat16,
+        stage_kwargs=dict(record_forward_time=False),
+    )
+    il_loss = engine.run_iteration(inputs, labels, num_microbatches=4)
+    assert abs(il_loss - ref_loss) < 0.05, (il_loss, ref_loss)
+
+    opt = FusedSGD(engine.parameters(), lr=1e-2)
+    losses = []
+    for _ in range(8):
+        opt.zero_grad()
+        losses.append(engine.run_iteration(inputs, labels, num_microbatches=4))
+        opt.step()
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0] * 0.95, losses
