@@ -145,11 +145,23 @@ DEV unsigned keep_to_16(float keep) {
 }
 
 // erf-formula GELU and its derivative (fp32)
-DEV float gelu_f(float x) { return 0.5f * x * (1.f + erff(x * 0.70710678118654752f)); }
+// Branch-free erf (Abramowitz & Stegun 7.1.26, |err| <= 1.5e-7 absolute —
+// far below bf16 resolution): ~12 VALU ops vs OCML erff's ~30 with range
+// branches. Feeds the exact (erf) GELU, not the tanh approximation.
+DEV float erf_fast(float x) {
+  const float ax = fabsf(x);
+  const float t = 1.f / fmaf(0.3275911f, ax, 1.f);
+  const float p = t * fmaf(t, fmaf(t, fmaf(t, fmaf(t, 1.061405429f, -1.453152027f),
+                                           1.421413741f), -0.284496736f),
+                           0.254829592f);
+  const float r = 1.f - p * __expf(-ax * ax);
+  return copysignf(r, x);
+}
+DEV float gelu_f(float x) { return 0.5f * x * (1.f + erf_fast(x * 0.70710678118654752f)); }
 DEV float gelu_grad_f(float x) {
   const float k = 0.70710678118654752f;       // 1/sqrt(2)
   const float c = 0.3989422804014327f;        // 1/sqrt(2*pi)
-  float cdf = 0.5f * (1.f + erff(x * k));
+  float cdf = 0.5f * (1.f + erf_fast(x * k));
   float pdf = c * __expf(-0.5f * x * x);
   return cdf + x * pdf;
 }
