@@ -121,10 +121,15 @@ def test_slowdown_injection_cpu():
         torch.ones(2, 8, dtype=torch.long),
     )
     slow(*args)  # warmup (lazy kernel init noise)
-    slow.reset_timing()
-    t0 = time.perf_counter(); slow(*args); wall = time.perf_counter() - t0
-    comp = slow.total_forward_time()
     # self-normalized: the injected sleep is 3x the stage's own measured
-    # compute, so wall >= ~comp * (1 + 3) regardless of machine speed
+    # compute, so wall >= ~comp * (1 + 3) regardless of machine speed; one
+    # re-measure tolerates a scheduling hiccup on a loaded machine
+    wall = comp = 0.0
+    for _attempt in range(2):
+        slow.reset_timing()
+        t0 = time.perf_counter(); slow(*args); wall = time.perf_counter() - t0
+        comp = slow.total_forward_time()
+        if comp > 0 and wall > comp * 2.5:
+            break
     assert comp > 0
     assert wall > comp * 2.5, (wall, comp)
