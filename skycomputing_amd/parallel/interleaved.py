@@ -114,7 +114,7 @@ class InterleavedPipelineEngine:
                 self.chunks[s] = build_module_from_cfg(
                     layer_cfgs[a:b], dtype=dtype, **kw
                 )
-        self._warmed = False
+        self._warmed: set = set()  # microbatch counts already handshaken
         self._orders: dict = {}
 
     def parameters(self):
@@ -128,10 +128,11 @@ class InterleavedPipelineEngine:
         (each peer's sends to me, in that peer's execution order — messages
         between a pair are untagged FIFO, so the receiver must drain them
         in the sender's order and stash out-of-order arrivals)."""
-        key = (M, self._warmed)
+        warmed = M in self._warmed
+        key = (M, warmed)
         if key not in self._orders:
             full = interleaved_schedule(self.owner, M,
-                                        serialized=not self._warmed)
+                                        serialized=not warmed)
             me = self.comm.rank
             inbound: dict = {}
             for peer, evs in full.items():
@@ -162,6 +163,7 @@ class InterleavedPipelineEngine:
         pending_sends = []
         total_loss = 0.0
 
+        warmed = M in self._warmed
         my_events, inbound_seq = self._schedule(M)
         fifo = {p: {"i": 0, "stash": {}} for p in inbound_seq}
 
@@ -208,7 +210,7 @@ class InterleavedPipelineEngine:
                             comm.isend_tensors(
                                 [t for t in out if torch.is_tensor(t)],
                                 self.owner[s + 1], f"if{s + 1}",
-                                blocking=not self._warmed,
+                                blocking=not warmed,
                             )
                         )
             else:  # backward
@@ -237,12 +239,12 @@ class InterleavedPipelineEngine:
                         pending_sends.extend(
                             comm.isend_tensors(
                                 in_grads, self.owner[s - 1], f"ib{s - 1}",
-                                blocking=not self._warmed,
+                                blocking=not warmed,
                             )
                         )
 
         for w in pending_sends:
             w.wait()
-        self._warmed = True
+        self._warmed.add(M)
         loss_val = total_loss if me == last_owner else None
         return PipelineEngine._broadcast_loss(self, loss_val)
