@@ -53,11 +53,17 @@ class ParameterServer:
     # ---------------- collective gather/scatter ----------------
 
     def gather_from_engine(self, engine, comm):
-        """Collective: every rank contributes its stage's per-layer state
-        dicts; rank 0 fills the store. Call on ALL ranks."""
-        if engine.stage_idx is not None:
+        """Collective: every rank contributes its stage's (or, for the
+        interleaved engine, every owned chunk's) per-layer state dicts;
+        rank 0 fills the store. Call on ALL ranks."""
+        if hasattr(engine, "chunks"):  # InterleavedPipelineEngine
+            payload = [
+                (engine.plan.ranges[s][0], chunk.get_layer_state_dicts())
+                for s, chunk in engine.chunks.items()
+            ]
+        elif engine.stage_idx is not None:
             start, _end = engine.plan.ranges[engine.stage_idx]
-            payload = (start, engine.stage.get_layer_state_dicts())
+            payload = [(start, engine.stage.get_layer_state_dicts())]
         else:
             payload = None
         gathered = comm.gather_object(payload, dst=0)
@@ -65,9 +71,9 @@ class ParameterServer:
             for item in gathered:
                 if item is None:
                     continue
-                start, dicts = item
-                for off, sd in enumerate(dicts):
-                    self._layers[start + off] = sd
+                for start, dicts in item:
+                    for off, sd in enumerate(dicts):
+                        self._layers[start + off] = sd
 
     def scatter_to_engine(self, engine, comm):
         """Collective: rank 0 broadcasts the full layer list; each rank
@@ -76,6 +82,10 @@ class ParameterServer:
         layers = comm.broadcast_object(self._layers if comm.rank == 0 else None, src=0)
         if comm.rank == 0:
             self._layers = layers
-        if engine.stage_idx is not None:
+        if hasattr(engine, "chunks"):  # InterleavedPipelineEngine
+            for s, chunk in engine.chunks.items():
+                start, end = engine.plan.ranges[s]
+                chunk.load_layer_state_dicts(layers[start:end])
+        elif engine.stage_idx is not None:
             start, end = engine.plan.ranges[engine.stage_idx]
             engine.stage.load_layer_state_dicts(layers[start:end])
