@@ -136,6 +136,16 @@ def main():
     batches = list(loader)
 
     graphed = None
+    if use_cuda and sd == 0 and not args.no_graph and virtual > 1:
+        from skycomputing_amd.parallel.interleaved_graph import GraphedInterleavedStep
+
+        data0, labels0 = batches[0]
+        try:
+            graphed = GraphedInterleavedStep(engine, opt, M, list(data0), labels0)
+        except Exception as e:  # pragma: no cover - eager fallback
+            print(f"[bench] rank {rank}: interleaved graph capture failed "
+                  f"({e!r}); falling back to eager", flush=True)
+            graphed = None
     if (use_cuda and sd == 0 and not args.no_graph
             and args.schedule == "gpipe" and virtual == 1):
         data0, labels0 = batches[0]

@@ -268,3 +268,52 @@ def test_interleaved_engine_single_rank_v2():
         opt.step()
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0] * 0.95, losses
+
+
+def test_graphed_interleaved_single_rank_v2():
+    """GraphedInterleavedStep at world=1, v=2: the zero-copy local
+    boundary (aliased pool leaves) and per-chunk graph replay must
+    reproduce the eager interleaved loss and train."""
+    torch.manual_seed(7)
+    from skycomputing_amd.models import bert_pipeline_config
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import init_distributed
+    from skycomputing_amd.parallel.interleaved import (
+        InterleavedPipelineEngine, build_interleaved_plan,
+    )
+    from skycomputing_amd.parallel.interleaved_graph import GraphedInterleavedStep
+
+    comm = init_distributed()
+    cfg = dict(hidden_size=256, num_attention_heads=4, intermediate_size=1024,
+               vocab_size=2000, hidden_dropout_prob=0.0,
+               attention_probs_dropout_prob=0.0)
+    cfgs = bert_pipeline_config(2, cfg)
+    ids = torch.randint(0, 2000, (16, 32))
+    inputs = (ids, torch.zeros_like(ids), torch.ones_like(ids))
+    labels = torch.randint(0, 3, (16,))
+    lf = lambda lg, lb: torch.nn.functional.cross_entropy(lg.float(), lb.to(lg.device))  # noqa: E731
+
+    torch.manual_seed(8)
+    plan = build_interleaved_plan(len(cfgs), 1, 2)
+    engine = InterleavedPipelineEngine(
+        comm, cfgs, plan, loss_fn=lf, dtype=torch.bfloat16,
+        stage_kwargs=dict(record_forward_time=False),
+    )
+    opt0 = FusedSGD(engine.parameters(), lr=0.0)
+    eager_loss = engine.run_iteration(inputs, labels, num_microbatches=4)
+    g = GraphedInterleavedStep(engine, opt0, 4, inputs, labels)
+    graphed_loss = g.step(inputs, labels)
+    assert abs(graphed_loss - eager_loss) < 0.05, (graphed_loss, eager_loss)
+    graphed_loss2 = g.step(inputs, labels)
+    assert abs(graphed_loss2 - graphed_loss) < 1e-4
+
+    torch.manual_seed(9)
+    engine2 = InterleavedPipelineEngine(
+        comm, cfgs, plan, loss_fn=lf, dtype=torch.bfloat16,
+        stage_kwargs=dict(record_forward_time=False),
+    )
+    opt = FusedSGD(engine2.parameters(), lr=1e-2)
+    g2 = GraphedInterleavedStep(engine2, opt, 4, inputs, labels)
+    losses = [g2.step(inputs, labels) for _ in range(10)]
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0] * 0.9, losses
