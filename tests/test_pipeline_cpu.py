@@ -267,3 +267,36 @@ def test_four_stage_gpipe_m8_matches_local(tmp_path):
     got = np.load(f"{tmp_path}/losses4.npy")
     assert np.allclose(got, np.array(ref), rtol=1e-4, atol=1e-5), (got, ref)
     assert got[-1] < got[0]
+
+
+def _isend_worker(rank, world_size, out_dir):
+    import torch as T
+
+    from skycomputing_amd.parallel import init_distributed
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    peer = 1 - rank
+    t1 = T.arange(8, dtype=T.float32)
+    t2 = T.full((3, 2), float(rank), dtype=T.float32)
+    # handshake (blocking) then non-blocking steady state
+    if rank == 0:
+        comm.isend_tensors([t1, t2], peer, "u", blocking=True)
+        works = comm.isend_tensors([t1 * 2, t2 * 2], peer, "u")
+        for w in works:
+            w.wait()
+    else:
+        a = comm.recv_tensors(peer, "u")
+        b = comm.recv_tensors(peer, "u")
+        assert T.equal(a[0], T.arange(8, dtype=T.float32))
+        assert T.equal(b[0], T.arange(8, dtype=T.float32) * 2)
+        assert a[1].shape == (3, 2) and float(b[1][0, 0]) == 0.0
+    comm.barrier()
+    from skycomputing_amd.parallel import destroy
+
+    destroy()
+
+
+def test_isend_tensors_transport():
+    """isend_tensors: blocking handshake first, then meta-less
+    non-blocking sends with buffer-retaining work handles."""
+    run_multiprocess(_isend_worker, 2, 29810, ".")
