@@ -141,9 +141,9 @@ class InterleavedPipelineEngine:
                 seq = []
                 for kind, s, m in evs:
                     if kind == "F" and s < self.S - 1 and self.owner[s + 1] == me:
-                        seq.append((f"if{s + 1}", ("F", s + 1, m)))
+                        seq.append((f"if{s + 1}m{M}", ("F", s + 1, m)))
                     elif kind == "B" and s > 0 and self.owner[s - 1] == me:
-                        seq.append((f"ib{s - 1}", ("B", s - 1, m)))
+                        seq.append((f"ib{s - 1}m{M}", ("B", s - 1, m)))
                 if seq:
                     inbound[peer] = seq
             self._orders[key] = (full.get(me, []), inbound)
@@ -209,7 +209,7 @@ class InterleavedPipelineEngine:
                         pending_sends.extend(
                             comm.isend_tensors(
                                 [t for t in out if torch.is_tensor(t)],
-                                self.owner[s + 1], f"if{s + 1}",
+                                self.owner[s + 1], f"if{s + 1}m{M}",
                                 blocking=not warmed,
                             )
                         )
@@ -238,7 +238,7 @@ class InterleavedPipelineEngine:
                     else:
                         pending_sends.extend(
                             comm.isend_tensors(
-                                in_grads, self.owner[s - 1], f"ib{s - 1}",
+                                in_grads, self.owner[s - 1], f"ib{s - 1}m{M}",
                                 blocking=not warmed,
                             )
                         )

@@ -79,7 +79,7 @@ class GraphedInterleavedStep:
                         for t in raw
                     ]
                 elif owner[s - 1] != me:
-                    metas = comm.cached_recv_meta(owner[s - 1], f"if{s}")
+                    metas = comm.cached_recv_meta(owner[s - 1], f"if{s}m{M}")
                     assert metas is not None, f"channel if{s} not handshaken"
                     self.static_in[(s, m)] = [
                         torch.empty(shape, dtype=dt, device=dev).requires_grad_(rq)
@@ -213,7 +213,7 @@ class GraphedInterleavedStep:
                 if s < S - 1 and owner[s + 1] != me:
                     comm.send_tensors(
                         [t for t in self.saved[(s, m)] if torch.is_tensor(t)],
-                        owner[s + 1], f"if{s + 1}",
+                        owner[s + 1], f"if{s + 1}m{M}",
                     )
             else:
                 if s < S - 1 and owner[s + 1] != me:
@@ -222,7 +222,7 @@ class GraphedInterleavedStep:
                 if s > 0 and owner[s - 1] != me:
                     in_grads = [t.grad for t in self.static_in[(s, m)]
                                 if torch.is_tensor(t) and t.requires_grad]
-                    comm.send_tensors(in_grads, owner[s - 1], f"ib{s - 1}")
+                    comm.send_tensors(in_grads, owner[s - 1], f"ib{s - 1}m{M}")
         self.g_opt.replay()
         loss = (float(self.static_loss.detach().cpu())
                 if me == self.last_owner else None)

@@ -207,3 +207,37 @@ def test_interleaved_checkpoint_roundtrip(tmp_path):
     multiple chunks per rank (save on rank 0, restore into a
     differently-initialized engine)."""
     run_multiprocess(_ckpt_worker, 2, 29790, tiny_bert_cfg(2), str(tmp_path))
+
+
+def _multi_m_worker(rank, world_size, layer_cfgs, batch, labels, out_dir):
+    torch.manual_seed(1234)
+    from skycomputing_amd.parallel import init_distributed
+    from skycomputing_amd.parallel.interleaved import (
+        InterleavedPipelineEngine, build_interleaved_plan,
+    )
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    plan = build_interleaved_plan(len(layer_cfgs), world_size, 2)
+    engine = InterleavedPipelineEngine(
+        comm, layer_cfgs, plan, loss_fn=torch.nn.CrossEntropyLoss(),
+        stage_kwargs=dict(record_forward_time=False),
+    )
+    # each microbatch count gets its own serialized handshake iteration;
+    # alternating Ms must not wedge or corrupt channels
+    losses = []
+    for M in (4, 2, 4, 2):
+        losses.append(engine.run_iteration(batch, labels, num_microbatches=M))
+    assert all(np.isfinite(losses)), losses
+    # same weights, same data: the two M=4 losses must match exactly
+    assert abs(losses[0] - losses[2]) < 1e-6
+    assert abs(losses[1] - losses[3]) < 1e-6
+    comm.barrier()
+    from skycomputing_amd.parallel import destroy
+
+    destroy()
+
+
+def test_interleaved_changing_microbatch_count(tmp_path):
+    batch, labels = _make_batch()
+    run_multiprocess(_multi_m_worker, 2, 29830, tiny_bert_cfg(2), batch,
+                     labels, str(tmp_path))
