@@ -300,3 +300,62 @@ def test_isend_tensors_transport():
     """isend_tensors: blocking handshake first, then meta-less
     non-blocking sends with buffer-retaining work handles."""
     run_multiprocess(_isend_worker, 2, 29810, ".")
+
+
+def _plan_worker(rank, world_size, layer_cfgs, batch, labels, lr, steps, M,
+                 schedule, ranges, out_dir):
+    torch.manual_seed(1234)
+    from skycomputing_amd.builder import build_module_from_cfg
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import PartitionPlan, PipelineEngine, init_distributed
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    full = build_module_from_cfg(layer_cfgs, record_forward_time=False)
+    plan = PartitionPlan(stage_ranks=list(range(len(ranges))), ranges=list(ranges))
+    engine = PipelineEngine(comm, layer_cfgs, plan, loss_fn=torch.nn.CrossEntropyLoss(),
+                            stage_kwargs=dict(record_forward_time=False))
+    if engine.stage_idx is not None:
+        start, end = plan.ranges[engine.stage_idx]
+        engine.stage.load_layer_state_dicts(
+            [
+                {k: v.detach().clone() for k, v in full.module[i].state_dict().items()}
+                for i in range(start, end)
+            ]
+        )
+    opt = FusedSGD(engine.parameters(), lr=lr)
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        losses.append(engine.run_iteration(batch, labels, num_microbatches=M,
+                                           schedule=schedule))
+        opt.step()
+    if rank == 0:
+        np.save(f"{out_dir}/losses_p.npy", np.array(losses, dtype=np.float64))
+    comm.barrier()
+    from skycomputing_amd.parallel import destroy
+
+    destroy()
+
+
+@pytest.mark.parametrize("case", ["lopsided_gpipe", "thin_middle_1f1b"])
+def test_adversarial_partitions_match_local(tmp_path, case):
+    """Extreme partitions the allocator can legitimately produce under
+    heterogeneity: a 1-layer first stage with everything else on stage 2,
+    and a 3-stage plan whose middle stage holds a single layer (1F1B)."""
+    layer_cfgs = tiny_bert_cfg(2)  # 9 layers
+    L = len(layer_cfgs)
+    batch, labels = _make_batch()
+    lr, steps = 0.05, 2
+    if case == "lopsided_gpipe":
+        world, M, schedule = 2, 8, "gpipe"
+        ranges = [(0, 1), (1, L)]
+        port = 29850
+    else:
+        world, M, schedule = 3, 4, "1f1b"
+        ranges = [(0, 4), (4, 5), (5, L)]
+        port = 29860
+    ref = _single_process_reference(layer_cfgs, batch, labels, lr, steps, M)
+    run_multiprocess(_plan_worker, world, port, layer_cfgs, batch, labels,
+                     lr, steps, M, schedule, ranges, str(tmp_path))
+    got = np.load(f"{tmp_path}/losses_p.npy")
+    assert np.allclose(got, np.array(ref), rtol=1e-4, atol=1e-5), (got, ref)
