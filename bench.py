@@ -98,7 +98,11 @@ def main():
         if rank == 0:
             workers = [dict(rank=r, **dev_results[r]) for r in range(world)]
             alloc = Allocator(model_results["flops"], model_results["mem"], workers)
-            plan = alloc.allocate(args.allocate)
+            v = args.virtual_stages if world > 1 else 1
+            if v > 1:
+                plan = alloc.interleaved_allocate(v)
+            else:
+                plan = alloc.allocate(args.allocate)
             plan_d = plan.to_dict()
         else:
             plan_d = None
@@ -111,7 +115,9 @@ def main():
             InterleavedPipelineEngine, build_interleaved_plan,
         )
 
-        plan = build_interleaved_plan(L, world, virtual)
+        if args.allocate == "even" and not args.stimulate:
+            plan = build_interleaved_plan(L, world, virtual)
+        # else: the benchmark-driven interleaved_allocate plan from above
         engine = InterleavedPipelineEngine(
             comm, layer_cfgs, plan,
             loss_fn=lambda logits, labels: torch.nn.functional.cross_entropy(

@@ -290,6 +290,111 @@ class Allocator:
             plan = self.refine_plan(plan)
         return plan
 
+    # ---------------- interleaved (virtual stages) ----------------
+
+    def device_cost(self, plan: PartitionPlan) -> float:
+        """max over devices of the SUM of their chunks' times — the
+        bottleneck metric for interleaved plans, where one device executes
+        several chunks per microbatch wave."""
+        by_rank = {w["rank"]: w for w in self.workers}
+        totals: dict = {}
+        for r, (a, b) in zip(plan.stage_ranks, plan.ranges):
+            totals[r] = totals.get(r, 0.0) + self._chunk_time(by_rank[r], a, b)
+        return max(totals.values())
+
+    def interleaved_allocate(self, v: int = 2) -> PartitionPlan:
+        """Heterogeneity-aware interleaved plan: v chunks per device in
+        round-robin order (d0..dk, d0..dk, ...), sized so each device's
+        TOTAL flops matches the exact min-max solution's share.
+
+        The per-device budget comes from ``optimal_allocate`` (merging a
+        device's chunks never changes its total, so the merged solver's
+        shares are optimal for the summed-bottleneck objective too); cuts
+        then split each share into v pieces along the round-robin device
+        cycle. Raises AllocationError if a device's chunks exceed its
+        memory (per-chunk memory tracks flops closely for uniform layer
+        stacks; heavily skewed memory profiles should fall back to
+        ``build_interleaved_plan``'s even split)."""
+        if v < 1:
+            raise ValueError("v must be >= 1")
+        base = self.optimal_allocate()
+        if v == 1:
+            return base
+        order = list(base.stage_ranks)
+        by_rank = {w["rank"]: w for w in self.workers}
+        share = {
+            r: self.F[b] - self.F[a]
+            for r, (a, b) in zip(base.stage_ranks, base.ranges)
+        }
+        # cut the layer list along the device cycle, each chunk taking
+        # ~1/v of its device's flop share (last chunk of the cycle absorbs
+        # rounding)
+        bounds = [0]
+        cycle = order * v
+        target = 0.0
+        for i, r in enumerate(cycle[:-1]):
+            target += share[r] / v
+            j = bisect_right(self.F, target)
+            j = max(bounds[-1], min(j - 1, self.L - (len(cycle) - 1 - i)))
+            bounds.append(j)
+        bounds.append(self.L)
+        bounds = self._balance_device_bounds(bounds, cycle)
+        plan = self._plan_from_bounds(bounds, cycle)
+        mem_used: dict = {}
+        for r, (a, b) in zip(plan.stage_ranks, plan.ranges):
+            mem_used[r] = mem_used.get(r, 0.0) + (self.M[b] - self.M[a])
+        for r, used in mem_used.items():
+            if used > by_rank[r]["avai_mem"] * (1 + 1e-9):
+                raise AllocationError(
+                    f"interleaved allocation exceeds rank {r} memory"
+                )
+        return plan
+
+    def _balance_device_bounds(self, bounds: list[int], cycle: list[int],
+                               break_iter: int = 2000) -> list[int]:
+        """Single-layer boundary shifts minimizing the per-DEVICE summed
+        bottleneck (repairs the layer-granularity rounding of the analytic
+        interleaved cuts; a one-layer shift moves flops between the two
+        devices adjacent in the cycle)."""
+        n = len(cycle)
+        by_rank = {w["rank"]: w for w in self.workers}
+
+        def totals(bs):
+            t: dict = {}
+            for i in range(n):
+                r = cycle[i]
+                t[r] = t.get(r, 0.0) + self._chunk_time(by_rank[r], bs[i], bs[i + 1])
+            return t
+
+        for _ in range(break_iter):
+            t = totals(bounds)
+            worst = max(t, key=t.get)
+            improved = False
+            for i in range(n):
+                if improved:
+                    break
+                if cycle[i] != worst:
+                    continue
+                for di in (-1, 1):
+                    j = i + di
+                    if not (0 <= j < n) or cycle[j] == worst:
+                        continue
+                    nb = list(bounds)
+                    if nb[i + 1] - nb[i] <= 0:
+                        continue
+                    if di == -1:
+                        nb[i] += 1      # first layer of i -> chunk j
+                    else:
+                        nb[i + 1] -= 1  # last layer of i -> chunk j
+                    nt = totals(nb)
+                    if max(nt[worst], nt[cycle[j]]) < t[worst] - 1e-12:
+                        bounds = nb
+                        improved = True
+                        break
+            if not improved:
+                break
+        return bounds
+
     def allocate(self, mode: str = "optimal") -> PartitionPlan:
         if mode == "even":
             return self.even_allocate()

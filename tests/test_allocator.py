@@ -229,3 +229,37 @@ def test_bert_boundary_payloads_shape():
     body_idx = [i for i, c in enumerate(cfgs) if c["layer_type"] == "BertLayer_Body"]
     for i in body_idx:
         assert pay[i] > pay[i - 1]
+
+
+# ---------------- interleaved (virtual-stage) allocation ----------------
+
+
+def test_interleaved_allocate_structure_and_cost():
+    """v chunks per device in round-robin order; the per-device summed
+    bottleneck must beat (or tie) the even interleaved split under
+    heterogeneous speeds."""
+    from skycomputing_amd.parallel.interleaved import build_interleaved_plan
+
+    a = _mk(L=48, W=4, times=[1.0, 3.0, 1.2, 2.0], seed=2)
+    p = a.interleaved_allocate(v=2)
+    _check_plan_cover(p, 48)
+    # round-robin device cycle, v appearances each
+    assert p.stage_ranks == a.optimal_allocate().stage_ranks * 2
+    even = build_interleaved_plan(48, 4, 2)
+    assert a.device_cost(p) <= a.device_cost(even) + 1e-9
+    # and lands within layer-granularity slack of the merged optimum's
+    # bottleneck (the per-device totals cannot always be hit exactly on
+    # whole-layer cuts)
+    assert a.device_cost(p) <= a.plan_cost(a.optimal_allocate()) * 1.10
+
+
+def test_interleaved_allocate_v1_is_optimal():
+    a = _mk(L=20, W=3, times=[1.0, 2.0, 1.5])
+    assert a.interleaved_allocate(v=1).ranges == a.optimal_allocate().ranges
+
+
+def _check_plan_cover(plan, L):
+    assert plan.ranges[0][0] == 0
+    assert plan.ranges[-1][1] == L
+    for (s, e), (s2, e2) in zip(plan.ranges, plan.ranges[1:]):
+        assert e == s2 and s < e and s2 < e2
