@@ -234,3 +234,33 @@ def test_chrome_trace_export(tmp_path):
     ev = trace["traceEvents"]
     assert len(ev) == 2
     assert all(e["ph"] == "X" and e["tid"] == 3 and e["dur"] > 0 for e in ev)
+
+
+def test_metrics_hook_writes_jsonl(tmp_path):
+    import json
+
+    from skycomputing_amd.builder import build_hook
+    from skycomputing_amd.runner import MetricsHook
+
+    path = str(tmp_path / "metrics.jsonl")
+    hook = build_hook(dict(layer_type="MetricsHook", path=path))
+    assert isinstance(hook, MetricsHook)
+
+    class R:  # minimal runner surface
+        comm = None
+        epoch = 0
+        iter = 0
+        last_loss = 1.25
+        iter_times = [0.01]
+
+    r = R()
+    hook.before_run(r)
+    hook.after_train_iter(r)
+    r.iter = 1
+    r.last_loss = 1.1
+    hook.after_train_iter(r)
+    hook.after_run(r)
+    recs = [json.loads(l) for l in open(path)]
+    assert len(recs) == 2
+    assert recs[0]["loss"] == 1.25 and recs[1]["iter"] == 1
+    assert recs[0]["iter_time_s"] == 0.01
