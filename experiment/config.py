@@ -71,6 +71,7 @@ train_config = dict(
         dict(layer_type="TimerHook"),
         dict(layer_type="StopHook", root="."),
         # dict(layer_type="CheckpointHook", save_path="./checkpoints", save_interval=1),
+        # dict(layer_type="MetricsHook", path="./metrics.jsonl"),
     ],
 )
 

@@ -89,3 +89,19 @@ def test_bench_virtual_stages_cpu(tmp_path):
     d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][0])
     assert d["config"]["parallelism"] == "pp4x2"
     assert d["value"] > 0
+
+
+def test_bench_virtual_stages_optimal_allocation(tmp_path):
+    """--virtual-stages combined with benchmark-driven allocation routes
+    through Allocator.interleaved_allocate (heterogeneity-aware chunk
+    sizing) end to end."""
+    out = tmp_path / "bvo.json"
+    r = _torchrun_bench(2, ["--gpus", "2", "--layers", "4", "--batch", "8",
+                            "--seq", "16", "--steps", "1", "--warmup", "1",
+                            "--virtual-stages", "2", "--allocate", "optimal",
+                            "--stimulate"], out)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][0])
+    assert d["config"]["parallelism"] == "pp2x2"
+    assert d["config"]["allocate"] == "optimal"
+    assert d["value"] > 0
