@@ -326,20 +326,41 @@ class Allocator:
             r: self.F[b] - self.F[a]
             for r, (a, b) in zip(base.stage_ranks, base.ranges)
         }
-        # cut the layer list along the device cycle, each chunk taking
-        # ~1/v of its device's flop share (last chunk of the cycle absorbs
-        # rounding)
-        bounds = [0]
-        cycle = order * v
-        target = 0.0
-        for i, r in enumerate(cycle[:-1]):
-            target += share[r] / v
-            j = bisect_right(self.F, target)
-            j = max(bounds[-1], min(j - 1, self.L - (len(cycle) - 1 - i)))
-            bounds.append(j)
-        bounds.append(self.L)
-        bounds = self._balance_device_bounds(bounds, cycle)
-        plan = self._plan_from_bounds(bounds, cycle)
+        # candidate starting points — analytic share-proportional cuts and
+        # the plain even split, over both the optimal-order cycle and the
+        # rank-order cycle — each polished by the device-cost local search;
+        # keep the best (the search only accepts improvements, so the
+        # result never loses to the plain even interleave)
+        def analytic_bounds(cycle):
+            S = len(cycle)
+            bs = [0]
+            target = 0.0
+            for i, r in enumerate(cycle[:-1]):
+                target += share.get(r, self.F[self.L] / len(share)) / v
+                j = bisect_right(self.F, target)
+                j = max(bs[-1], min(j - 1, self.L - (S - 1 - i)))
+                bs.append(j)
+            bs.append(self.L)
+            return bs
+
+        def even_bounds(cycle):
+            S = len(cycle)
+            base_c, rem_c = divmod(self.L, S)
+            bs = [0]
+            for i in range(S):
+                bs.append(bs[-1] + base_c + (1 if i < rem_c else 0))
+            return bs
+
+        cycles = [order * v]
+        rank_cycle = [w["rank"] for w in self.workers if w["rank"] in share] * v
+        if rank_cycle != cycles[0]:
+            cycles.append(rank_cycle)
+        plans = []
+        for cycle in cycles:
+            for bs in (analytic_bounds(cycle), even_bounds(cycle)):
+                bb = self._balance_device_bounds(bs, cycle)
+                plans.append(self._plan_from_bounds(bb, cycle))
+        plan = min(plans, key=self.device_cost)
         mem_used: dict = {}
         for r, (a, b) in zip(plan.stage_ranks, plan.ranges):
             mem_used[r] = mem_used.get(r, 0.0) + (self.M[b] - self.M[a])
