@@ -261,7 +261,11 @@ SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
   hipStream_t s = (hipStream_t)stream;
   int64_t n = rows * cols;
   const bool hb = b != 0;
-  if (cols % 8 == 0 && scratch != 0) {
+  // The fused one-pass variant measures FASTER standalone but SLOWER
+  // in-app (+4.9 us/layer vs the dx + two-stage-colsum chain at the bench
+  // shape — profiles r01_bench160 diff); the proven chain is the default
+  // and the fused kernel stays selectable for re-measurement.
+  if (cols % 8 == 0 && scratch != 0 && getenv("SKY_GELU_FUSED_BWD")) {
     // fused one-pass dx + db partials (see bias_gelu_bwd_part_kernel).
     // This kernel moves 3 tensors (dy, x read; dx write), so unlike the
     // read-only colsum part it wants FULL thread count (~4096 waves, short
