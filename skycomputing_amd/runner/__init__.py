@@ -1,5 +1,7 @@
 from .hooks import Hook
-from .hooks_collection import CheckpointHook, MetricsHook, StopHook, TimerHook
+from .hooks_collection import (
+    CheckpointHook, LRScheduleHook, MetricsHook, StopHook, TimerHook,
+)
 from .runner import Runner
 
-__all__ = ["Runner", "Hook", "CheckpointHook", "MetricsHook", "StopHook", "TimerHook"]
+__all__ = ["Runner", "Hook", "CheckpointHook", "LRScheduleHook", "MetricsHook", "StopHook", "TimerHook"]

@@ -1,6 +1,7 @@
 from .checkpoint_hook import CheckpointHook
+from .lr_hook import LRScheduleHook
 from .metrics_hook import MetricsHook
 from .stop_hook import StopHook
 from .timer_hook import TimerHook
 
-__all__ = ["CheckpointHook", "MetricsHook", "StopHook", "TimerHook"]
+__all__ = ["CheckpointHook", "LRScheduleHook", "MetricsHook", "StopHook", "TimerHook"]

@@ -30,6 +30,7 @@ class Runner:
         schedule: str = "gpipe",
         logger: Logger | None = None,
         log_interval: int = 1,
+        grad_clip_norm: float | None = None,
     ):
         self.engine = engine
         self.optimizer = optimizer
@@ -40,6 +41,7 @@ class Runner:
         self.schedule = schedule
         self.logger = logger or Logger(rank=comm.rank)
         self.log_interval = log_interval
+        self.grad_clip_norm = grad_clip_norm
         self.timer = DistributedTimer()
         self.hooks: list[Hook] = []
         self.epoch = 0
@@ -79,6 +81,14 @@ class Runner:
             num_microbatches=self.num_microbatches,
             schedule=self.schedule,
         )
+        if self.grad_clip_norm is not None:
+            # per-rank clip over this stage's grads (each rank owns its
+            # slice; the reference has no clipping at all)
+            import torch
+
+            torch.nn.utils.clip_grad_norm_(
+                self.engine.parameters(), self.grad_clip_norm
+            )
         self.optimizer.step()
         self.iter_times.append(time.perf_counter() - t0)
         self.last_loss = loss

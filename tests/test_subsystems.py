@@ -264,3 +264,33 @@ def test_metrics_hook_writes_jsonl(tmp_path):
     assert len(recs) == 2
     assert recs[0]["loss"] == 1.25 and recs[1]["iter"] == 1
     assert recs[0]["iter_time_s"] == 0.01
+
+
+def test_lr_schedule_hook_math():
+    from skycomputing_amd.builder import build_hook
+    from skycomputing_amd.runner import LRScheduleHook
+
+    h = build_hook(dict(layer_type="LRScheduleHook", base_lr=1.0,
+                        warmup_iters=10, total_iters=110, decay="linear",
+                        min_lr=0.1))
+    assert isinstance(h, LRScheduleHook)
+    assert h.lr_at(0, 110) == pytest.approx(0.1)   # warmup ramp start
+    assert h.lr_at(9, 110) == pytest.approx(1.0)
+    assert h.lr_at(10, 110) == pytest.approx(1.0)
+    mid = h.lr_at(60, 110)
+    assert 0.1 < mid < 1.0
+    assert h.lr_at(109, 110) == pytest.approx(0.1, abs=0.02)
+    hc = LRScheduleHook(base_lr=2.0, warmup_iters=0, total_iters=100, decay="cosine")
+    assert hc.lr_at(0, 100) == pytest.approx(2.0)
+    assert hc.lr_at(99, 100) == pytest.approx(0.0, abs=0.002)
+
+    class Opt:
+        lr = 0.0
+
+    class R:
+        iter = 5
+        max_iter = 110
+        optimizer = Opt()
+
+    h.before_train_iter(R)
+    assert R.optimizer.lr == pytest.approx(0.6)
