@@ -26,11 +26,16 @@ class CheckpointHook(Hook):
         save_interval: int = 1,
         load_from: str | None = None,
         num_layers: int | None = None,
+        resume_counters: bool = False,
     ):
+        """``resume_counters``: also restore runner.epoch/iter from the
+        checkpoint meta (continue the schedule where it stopped) instead
+        of restarting the counters with restored weights."""
         self.save_path = save_path
         self.save_interval = save_interval
         self.load_from = load_from
         self.num_layers = num_layers
+        self.resume_counters = resume_counters
         self._ps: ParameterServer | None = None
 
     def _server(self, runner) -> ParameterServer:
@@ -43,9 +48,14 @@ class CheckpointHook(Hook):
         if not self.load_from:
             return
         ps = self._server(runner)
+        meta = None
         if runner.comm.rank == 0:
-            ps.load_weights_from_file(self.load_from)
+            meta = ps.load_weights_from_file(self.load_from)
         ps.scatter_to_engine(runner.engine, runner.comm)
+        if self.resume_counters:
+            meta = runner.comm.broadcast_object(meta, src=0) or {}
+            runner.epoch = int(meta.get("epoch", 0))
+            runner.iter = int(meta.get("iter", 0))
         runner.logger.info(f"restored checkpoint from {self.load_from}")
 
     def after_train_epoch(self, runner):

@@ -98,7 +98,9 @@ class Runner:
         self.engine.train(True)
         self.call_hook("before_run")
         done = False
-        for epoch in range(self._max_epoch):
+        # start from self.epoch so a resumed checkpoint (CheckpointHook
+        # resume_counters) continues its schedule
+        for epoch in range(self.epoch, self._max_epoch):
             self.epoch = epoch
             self.call_hook("before_train_epoch")
             for data, labels in data_loader:

@@ -294,3 +294,14 @@ def test_lr_schedule_hook_math():
 
     h.before_train_iter(R)
     assert R.optimizer.lr == pytest.approx(0.6)
+
+
+def test_checkpoint_meta_roundtrip(tmp_path):
+    """Checkpoint meta carries epoch/iter for counter resume."""
+    ps = ParameterServer(1)
+    ps.update_weights({"w": torch.randn(2)}, 0)
+    path = str(tmp_path / "epoch_3.pth")
+    ps.save_weights_to_file(path, meta={"epoch": 3, "iter": 90})
+    ps2 = ParameterServer(1)
+    meta = ps2.load_weights_from_file(path)
+    assert meta["epoch"] == 3 and meta["iter"] == 90
