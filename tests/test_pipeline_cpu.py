@@ -359,3 +359,47 @@ def test_adversarial_partitions_match_local(tmp_path, case):
                      lr, steps, M, schedule, ranges, str(tmp_path))
     got = np.load(f"{tmp_path}/losses_p.npy")
     assert np.allclose(got, np.array(ref), rtol=1e-4, atol=1e-5), (got, ref)
+
+
+def _resume_worker(rank, world_size, layer_cfgs, out_dir):
+    torch.manual_seed(0)
+    from skycomputing_amd.builder import build_dataloader_from_cfg
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import PartitionPlan, PipelineEngine, init_distributed
+    from skycomputing_amd.runner import CheckpointHook, Runner
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    loader = build_dataloader_from_cfg(
+        4, dict(type="SyntheticGlueDataset", size=16, max_seq_length=8,
+                vocab_size=500, num_class=3, seed=3))
+    plan = PartitionPlan(stage_ranks=[0], ranges=[(0, len(layer_cfgs))])
+
+    def make_runner(max_epoch, hook):
+        eng = PipelineEngine(comm, layer_cfgs, plan,
+                             loss_fn=torch.nn.CrossEntropyLoss(),
+                             stage_kwargs=dict(record_forward_time=False))
+        opt = FusedSGD(eng.parameters(), lr=0.01)
+        r = Runner(eng, opt, comm, max_epoch=max_epoch)
+        r.register_hook(hook)
+        return r
+
+    # phase 1: train 2 epochs, checkpoint each
+    r1 = make_runner(2, CheckpointHook(save_path=f"{out_dir}/ck", save_interval=1))
+    r1.train(loader)
+    assert r1.epoch == 1 and r1.iter == 8  # 2 epochs x 4 batches
+
+    # phase 2: resume from epoch_1 with counters -> trains ONLY epoch 2
+    r2 = make_runner(2, CheckpointHook(
+        save_path=f"{out_dir}/ck2", save_interval=1,
+        load_from=f"{out_dir}/ck/epoch_1.pth", resume_counters=True))
+    r2.train(loader)
+    assert r2.epoch == 1
+    assert r2.iter == 4 + 4  # resumed at iter 4, ran one more epoch
+    comm.barrier()
+    from skycomputing_amd.parallel import destroy
+
+    destroy()
+
+
+def test_runner_resume_counters(tmp_path):
+    run_multiprocess(_resume_worker, 1, 29870, tiny_bert_cfg(1), str(tmp_path))
