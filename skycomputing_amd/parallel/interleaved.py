@@ -248,3 +248,39 @@ class InterleavedPipelineEngine:
         self._warmed.add(M)
         loss_val = total_loss if me == last_owner else None
         return PipelineEngine._broadcast_loss(self, loss_val)
+
+    def evaluate_batch(self, data, labels=None):
+        """Forward-only pass through the chunk chain (whole batch, no
+        microbatching); returns logits on the last-chunk owner, None
+        elsewhere. Eval uses its own always-handshaking channels — output
+        requires_grad flags differ from training, so the channels must not
+        be shared (same rule as PipelineEngine's fwd_eval key)."""
+        comm, S = self.comm, self.S
+        me = comm.rank
+        for c in self.chunks.values():
+            c.eval()
+        try:
+            with torch.no_grad():
+                cur = None
+                for s in range(S):
+                    if self.owner[s] != me:
+                        continue
+                    if s == 0:
+                        ins = list(data) if isinstance(data, (tuple, list)) else [data]
+                    elif self.owner[s - 1] == me:
+                        ins = cur
+                    else:
+                        ins = comm.recv_tensors(self.owner[s - 1], f"iev{s}")
+                    out = self.chunks[s](*ins)
+                    cur = list(out) if isinstance(out, (tuple, list)) else [out]
+                    if s < S - 1 and self.owner[s + 1] != me:
+                        comm.send_tensors(
+                            [t for t in cur if torch.is_tensor(t)],
+                            self.owner[s + 1], f"iev{s + 1}",
+                        )
+        finally:
+            for c in self.chunks.values():
+                c.train()
+        if me == self.owner[-1]:
+            return cur[0]
+        return None

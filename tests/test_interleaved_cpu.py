@@ -241,3 +241,38 @@ def test_interleaved_changing_microbatch_count(tmp_path):
     batch, labels = _make_batch()
     run_multiprocess(_multi_m_worker, 2, 29830, tiny_bert_cfg(2), batch,
                      labels, str(tmp_path))
+
+
+def _eval_worker(rank, world_size, layer_cfgs, batch, out_dir):
+    torch.manual_seed(11)
+    from skycomputing_amd.parallel import init_distributed
+    from skycomputing_amd.parallel.interleaved import (
+        InterleavedPipelineEngine, build_interleaved_plan,
+    )
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    plan = build_interleaved_plan(len(layer_cfgs), world_size, 2)
+    engine = InterleavedPipelineEngine(
+        comm, layer_cfgs, plan, loss_fn=torch.nn.CrossEntropyLoss(),
+        stage_kwargs=dict(record_forward_time=False),
+    )
+    logits = engine.evaluate_batch(batch)
+    last_owner = plan.stage_ranks[-1]
+    if rank == last_owner:
+        assert logits is not None and logits.shape == (8, 3)
+        assert torch.isfinite(logits).all()
+    else:
+        assert logits is None
+    # a second eval reuses the channels cleanly
+    logits2 = engine.evaluate_batch(batch)
+    if rank == last_owner:
+        assert torch.allclose(logits, logits2)
+    comm.barrier()
+    from skycomputing_amd.parallel import destroy
+
+    destroy()
+
+
+def test_interleaved_evaluate_batch(tmp_path):
+    batch, _labels = _make_batch()
+    run_multiprocess(_eval_worker, 2, 29880, tiny_bert_cfg(2), batch, str(tmp_path))
