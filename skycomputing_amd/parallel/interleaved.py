@@ -123,6 +123,14 @@ class InterleavedPipelineEngine:
             out.extend(c.parameters())
         return out
 
+    def train(self, mode: bool = True):
+        for c in self.chunks.values():
+            c.train(mode)
+        return self
+
+    def eval(self):
+        return self.train(False)
+
     def _schedule(self, M: int):
         """Full per-rank event lists + my inbound message sequence per peer
         (each peer's sends to me, in that peer's execution order — messages
@@ -150,7 +158,13 @@ class InterleavedPipelineEngine:
         return self._orders[key]
 
 
-    def run_iteration(self, data, labels, num_microbatches: int = 1):
+    def run_iteration(self, inputs=None, labels=None,
+                      num_microbatches: int = 1,
+                      schedule: str | None = None):
+        """Same signature as PipelineEngine.run_iteration; ``schedule`` is
+        accepted for Runner compatibility and ignored — the interleaved
+        engine's event order IS its schedule."""
+        data = inputs
         comm, S, M = self.comm, self.S, num_microbatches
         me = comm.rank
         first_owner, last_owner = self.owner[0], self.owner[-1]

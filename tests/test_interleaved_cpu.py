@@ -276,3 +276,40 @@ def _eval_worker(rank, world_size, layer_cfgs, batch, out_dir):
 def test_interleaved_evaluate_batch(tmp_path):
     batch, _labels = _make_batch()
     run_multiprocess(_eval_worker, 2, 29880, tiny_bert_cfg(2), batch, str(tmp_path))
+
+
+def _runner_worker(rank, world_size, layer_cfgs, out_dir):
+    torch.manual_seed(21)
+    from skycomputing_amd.builder import build_dataloader_from_cfg
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import init_distributed
+    from skycomputing_amd.parallel.interleaved import (
+        InterleavedPipelineEngine, build_interleaved_plan,
+    )
+    from skycomputing_amd.runner import Runner
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    plan = build_interleaved_plan(len(layer_cfgs), world_size, 2)
+    engine = InterleavedPipelineEngine(
+        comm, layer_cfgs, plan, loss_fn=torch.nn.CrossEntropyLoss(),
+        stage_kwargs=dict(record_forward_time=False),
+    )
+    opt = FusedSGD(engine.parameters(), lr=0.02)
+    loader = build_dataloader_from_cfg(
+        4, dict(type="SyntheticGlueDataset", size=12, max_seq_length=8,
+                vocab_size=500, num_class=3, seed=9))
+    r = Runner(engine, opt, comm, max_epoch=2, num_microbatches=2)
+    r.train(loader)
+    assert r.last_loss is not None and np.isfinite(r.last_loss)
+    acc = r.val(loader, max_batches=2)
+    assert acc is not None and 0.0 <= acc <= 1.0
+    comm.barrier()
+    from skycomputing_amd.parallel import destroy
+
+    destroy()
+
+
+def test_runner_drives_interleaved_engine(tmp_path):
+    """The training Runner (hooks, val loop) works unchanged over the
+    interleaved engine — same engine interface surface."""
+    run_multiprocess(_runner_worker, 2, 29890, tiny_bert_cfg(2), str(tmp_path))
