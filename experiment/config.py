@@ -57,6 +57,9 @@ allocator_config = dict(
     mode=os.environ.get("ALLOCATE_TYPE", "even"),  # even | dynamic | optimal
     benchmark=dict(batch_size=32, seq_len=128, iterations=5),
     stimulate=os.environ.get("STIMULATE") == "1",
+    # >1: interleaved virtual stages (v chunks per rank; heterogeneity-aware
+    # sizing when mode != even)
+    virtual_stages=int(os.environ.get("SKY_VIRTUAL_STAGES", "1")),
 )
 
 # ---- training (reference config.py:154-167) ----

@@ -88,15 +88,21 @@ def main():
 
     acfg = cfg.get("allocator_config", {"mode": "even"})
     mode = args.allocate or acfg.get("mode", "even")
+    virtual = int(acfg.get("virtual_stages", 1)) if world > 1 else 1
     if mode == "even" and not acfg.get("stimulate"):
-        base, rem = divmod(L, world)
-        bounds = [0]
-        for i in range(world):
-            bounds.append(bounds[-1] + base + (1 if i < rem else 0))
-        plan = PartitionPlan(
-            stage_ranks=list(range(world)),
-            ranges=[(bounds[i], bounds[i + 1]) for i in range(world)],
-        )
+        if virtual > 1:
+            from skycomputing_amd.parallel.interleaved import build_interleaved_plan
+
+            plan = build_interleaved_plan(L, world, virtual)
+        else:
+            base, rem = divmod(L, world)
+            bounds = [0]
+            for i in range(world):
+                bounds.append(bounds[-1] + base + (1 if i < rem else 0))
+            plan = PartitionPlan(
+                stage_ranks=list(range(world)),
+                ranges=[(bounds[i], bounds[i + 1]) for i in range(world)],
+            )
     else:
         bench = acfg.get("benchmark", {})
         stim = Stimulator(world) if acfg.get("stimulate") else None
@@ -118,7 +124,10 @@ def main():
             model_results = mb.benchmark()
             workers = [dict(rank=r, **dev_results[r]) for r in range(world)]
             alloc = Allocator(model_results["flops"], model_results["mem"], workers)
-            plan = alloc.allocate(mode)
+            if virtual > 1:
+                plan = alloc.interleaved_allocate(virtual)
+            else:
+                plan = alloc.allocate(mode)
             logger.info(
                 f"allocation ({mode}): "
                 + ", ".join(f"r{r}:[{a},{b})" for r, (a, b) in zip(plan.stage_ranks, plan.ranges))
@@ -127,19 +136,33 @@ def main():
         else:
             plan_d = None
         plan = PartitionPlan.from_dict(comm.broadcast_object(plan_d, src=0))
-        for r, rng in zip(plan.stage_ranks, plan.ranges):
-            wm.assign_model_to_worker(r, rng)
+        if virtual == 1:
+            for r, rng in zip(plan.stage_ranks, plan.ranges):
+                wm.assign_model_to_worker(r, rng)
 
     sd = float(extra[rank].get("slowdown") or 0.0)
-    engine = PipelineEngine(
-        comm, layer_cfgs, plan,
-        loss_fn=lambda logits, labels: torch.nn.functional.cross_entropy(
-            logits.float(), labels
-        ),
-        dtype=dtype,
-        stage_kwargs=dict(record_forward_time=True, slowdown=sd,
-                          mem_limit=extra[rank].get("mem_limit")),
-    )
+    if virtual > 1:
+        from skycomputing_amd.parallel.interleaved import InterleavedPipelineEngine
+
+        engine = InterleavedPipelineEngine(
+            comm, layer_cfgs, plan,
+            loss_fn=lambda logits, labels: torch.nn.functional.cross_entropy(
+                logits.float(), labels
+            ),
+            dtype=dtype,
+            stage_kwargs=dict(record_forward_time=True, slowdown=sd,
+                              mem_limit=extra[rank].get("mem_limit")),
+        )
+    else:
+        engine = PipelineEngine(
+            comm, layer_cfgs, plan,
+            loss_fn=lambda logits, labels: torch.nn.functional.cross_entropy(
+                logits.float(), labels
+            ),
+            dtype=dtype,
+            stage_kwargs=dict(record_forward_time=True, slowdown=sd,
+                              mem_limit=extra[rank].get("mem_limit")),
+        )
 
     tcfg = cfg.train_config
     opt_cfg = tcfg.get("optimizer", {})

@@ -278,3 +278,25 @@ def test_baseline_configs_load():
         assert bool(cfg.allocator_config.get("stimulate")) == stim
         assert cfg.data_config["batch_size"] == 32
         assert cfg.train_config["max_iter"] == 30
+
+
+def test_launch_cli_virtual_stages(tmp_path):
+    """The canonical CLI with allocator_config.virtual_stages=2 routes
+    through the interleaved engine + interleaved_allocate end to end."""
+    cfg = tmp_path / "cfgv.py"
+    cfg.write_text(TINY_CONFIG.format(logdir=str(tmp_path / "logs"))
+                   .replace('stimulate=False,', 'stimulate=False,\n    virtual_stages=2,'))
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    r = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run", "--standalone",
+            "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+            os.path.join(REPO, "experiment", "launch.py"),
+            "-c", str(cfg),
+        ],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=400,
+    )
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    log = (tmp_path / "logs" / "rank0.log").read_text()
+    assert "done: 3 iterations" in log
