@@ -119,7 +119,8 @@ class GlueDataset(Dataset):
     }
 
     def __init__(self, data_dir: str, task: str = "mnli", vocab_file: str | None = None,
-                 max_seq_length: int = 128, split: str = "train"):
+                 max_seq_length: int = 128, split: str = "train",
+                 cache: bool = True):
         task = task.lower()
         if task not in self.TASK_LABELS:
             raise ValueError(f"unknown GLUE task {task}")
@@ -128,6 +129,20 @@ class GlueDataset(Dataset):
                 f"GLUE data dir {data_dir} not found; this image has no network — "
                 "use SyntheticGlueDataset for offline runs"
             )
+        # tokenized-feature cache next to the data (the reference pickled
+        # features too, scaelum/dataset/bert_dataset.py:42-66)
+        cache_path = os.path.join(
+            data_dir, f"cached_{task}_{split}_{max_seq_length}.pt"
+        )
+        if cache and os.path.isfile(cache_path):
+            import torch as _t
+
+            blob = _t.load(cache_path, weights_only=True)
+            self.input_ids = blob["input_ids"]
+            self.attention_mask = blob["attention_mask"]
+            self.token_type_ids = blob["token_type_ids"]
+            self.labels = blob["labels"]
+            return
         from transformers import BertTokenizerFast
 
         vocab = vocab_file or os.path.join(data_dir, "vocab.txt")
@@ -150,6 +165,17 @@ class GlueDataset(Dataset):
             "token_type_ids", torch.zeros_like(self.input_ids)
         )
         self.labels = torch.tensor([label_map[l] for _, _, l in examples], dtype=torch.long)
+        if cache:
+            try:
+                torch.save(
+                    dict(input_ids=self.input_ids,
+                         attention_mask=self.attention_mask,
+                         token_type_ids=self.token_type_ids,
+                         labels=self.labels),
+                    cache_path,
+                )
+            except OSError:
+                pass  # read-only data dir: skip caching
 
     @staticmethod
     def _read_examples(data_dir: str, task: str, split: str):

@@ -305,3 +305,20 @@ def test_checkpoint_meta_roundtrip(tmp_path):
     ps2 = ParameterServer(1)
     meta = ps2.load_weights_from_file(path)
     assert meta["epoch"] == 3 and meta["iter"] == 90
+
+
+def test_glue_dataset_feature_cache(tmp_path):
+    vocab = ["[PAD]", "[UNK]", "[CLS]", "[SEP]", "[MASK]", "the", "cat"]
+    (tmp_path / "vocab.txt").write_text("\n".join(vocab) + "\n")
+    (tmp_path / "train.tsv").write_text("sentence\tlabel\nthe cat\t1\n")
+    try:
+        ds1 = GlueDataset(str(tmp_path), task="sst-2", max_seq_length=8)
+    except ImportError:
+        pytest.skip("transformers unavailable")
+    cache = tmp_path / "cached_sst-2_train_8.pt"
+    assert cache.is_file()
+    # second load comes from the cache even without the tokenizer inputs
+    (tmp_path / "train.tsv").unlink()
+    ds2 = GlueDataset(str(tmp_path), task="sst-2", max_seq_length=8)
+    assert torch.equal(ds1.input_ids, ds2.input_ids)
+    assert torch.equal(ds1.labels, ds2.labels)
