@@ -27,15 +27,20 @@ class CheckpointHook(Hook):
         load_from: str | None = None,
         num_layers: int | None = None,
         resume_counters: bool = False,
+        save_best: bool = False,
     ):
         """``resume_counters``: also restore runner.epoch/iter from the
         checkpoint meta (continue the schedule where it stopped) instead
-        of restarting the counters with restored weights."""
+        of restarting the counters with restored weights.
+        ``save_best``: additionally write ``best.pth`` whenever a val
+        epoch improves ``runner.last_val_acc``."""
         self.save_path = save_path
         self.save_interval = save_interval
         self.load_from = load_from
         self.num_layers = num_layers
         self.resume_counters = resume_counters
+        self.save_best = save_best
+        self._best_acc: float | None = None
         self._ps: ParameterServer | None = None
 
     def _server(self, runner) -> ParameterServer:
@@ -68,3 +73,20 @@ class CheckpointHook(Hook):
             path = os.path.join(self.save_path, f"epoch_{runner.epoch + 1}.pth")
             ps.save_weights_to_file(path, meta={"epoch": runner.epoch + 1, "iter": runner.iter})
             runner.logger.info(f"saved checkpoint {path}")
+
+    def after_val_epoch(self, runner):
+        if not self.save_best:
+            return
+        acc = getattr(runner, "last_val_acc", None)
+        if acc is None or (self._best_acc is not None and acc <= self._best_acc):
+            return
+        self._best_acc = acc
+        ps = self._server(runner)
+        ps.gather_from_engine(runner.engine, runner.comm)
+        if runner.comm.rank == 0:
+            os.makedirs(self.save_path, exist_ok=True)
+            path = os.path.join(self.save_path, "best.pth")
+            ps.save_weights_to_file(
+                path, meta={"epoch": runner.epoch, "iter": runner.iter,
+                            "val_acc": acc})
+            runner.logger.info(f"saved best checkpoint {path} (acc {acc:.4f})")

@@ -47,6 +47,7 @@ class Runner:
         self.epoch = 0
         self.iter = 0
         self.last_loss: float | None = None
+        self.last_val_acc: float | None = None
         self.should_stop = False
         self.iter_times: list[float] = []
 
@@ -145,6 +146,7 @@ class Runner:
         acc = (correct / total) if total else None
         src = self.engine.plan.stage_ranks[-1] if self.engine.plan.num_stages else 0
         acc = self.comm.broadcast_object(acc, src=src)
+        self.last_val_acc = acc
         self.call_hook("after_val_epoch")
         self.engine.train(True)
         if acc is not None:
