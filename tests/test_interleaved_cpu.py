@@ -313,3 +313,34 @@ def test_runner_drives_interleaved_engine(tmp_path):
     """The training Runner (hooks, val loop) works unchanged over the
     interleaved engine — same engine interface surface."""
     run_multiprocess(_runner_worker, 2, 29890, tiny_bert_cfg(2), str(tmp_path))
+
+
+def _idle_worker(rank, world_size, layer_cfgs, batch, labels, out_dir):
+    torch.manual_seed(31)
+    from skycomputing_amd.parallel import PartitionPlan, init_distributed
+    from skycomputing_amd.parallel.interleaved import InterleavedPipelineEngine
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    L = len(layer_cfgs)
+    # rank 2 excluded (e.g. the allocator dropped a slow device)
+    plan = PartitionPlan(stage_ranks=[0, 1, 0, 1],
+                         ranges=[(0, 2), (2, 4), (4, 6), (6, L)])
+    engine = InterleavedPipelineEngine(
+        comm, layer_cfgs, plan, loss_fn=torch.nn.CrossEntropyLoss(),
+        stage_kwargs=dict(record_forward_time=False),
+    )
+    losses = [engine.run_iteration(batch, labels, num_microbatches=2)
+              for _ in range(2)]
+    assert all(np.isfinite(losses)), losses
+    if rank == 2:
+        assert not engine.chunks  # genuinely idle, still gets the loss
+    comm.barrier()
+    from skycomputing_amd.parallel import destroy
+
+    destroy()
+
+
+def test_interleaved_idle_rank(tmp_path):
+    batch, labels = _make_batch()
+    run_multiprocess(_idle_worker, 3, 29910, tiny_bert_cfg(2), batch, labels,
+                     str(tmp_path))
