@@ -286,7 +286,7 @@ def _runner_worker(rank, world_size, layer_cfgs, out_dir):
     from skycomputing_amd.parallel.interleaved import (
         InterleavedPipelineEngine, build_interleaved_plan,
     )
-    from skycomputing_amd.runner import Runner
+    from skycomputing_amd.runner import Runner, TimerHook
 
     comm = init_distributed(backend="gloo", timeout_s=60)
     plan = build_interleaved_plan(len(layer_cfgs), world_size, 2)
@@ -299,8 +299,15 @@ def _runner_worker(rank, world_size, layer_cfgs, out_dir):
         4, dict(type="SyntheticGlueDataset", size=12, max_seq_length=8,
                 vocab_size=500, num_class=3, seed=9))
     r = Runner(engine, opt, comm, max_epoch=2, num_microbatches=2)
+    r.register_hook(TimerHook(trace_path=f"{out_dir}/trace"))
     r.train(loader)
     assert r.last_loss is not None and np.isfinite(r.last_loss)
+    import json as _json
+    import os as _os
+
+    tr = f"{out_dir}/trace.rank{rank}.json"
+    assert _os.path.isfile(tr)
+    assert len(_json.load(open(tr))["traceEvents"]) > 0
     acc = r.val(loader, max_batches=2)
     assert acc is not None and 0.0 <= acc <= 1.0
     comm.barrier()
