@@ -443,3 +443,56 @@ def _best_ckpt_worker(rank, world_size, layer_cfgs, out_dir):
 
 def test_best_checkpoint_tracking(tmp_path):
     run_multiprocess(_best_ckpt_worker, 1, 29900, tiny_bert_cfg(1), str(tmp_path))
+
+
+def _four_stage_1f1b_worker(rank, world_size, layer_cfgs, batch, labels, lr,
+                            steps, M, out_dir):
+    torch.manual_seed(1234)
+    from skycomputing_amd.builder import build_module_from_cfg
+    from skycomputing_amd.optim import FusedSGD
+    from skycomputing_amd.parallel import PartitionPlan, PipelineEngine, init_distributed
+
+    comm = init_distributed(backend="gloo", timeout_s=60)
+    full = build_module_from_cfg(layer_cfgs, record_forward_time=False)
+    L = len(layer_cfgs)
+    base, rem = divmod(L, world_size)
+    bounds = [0]
+    for i in range(world_size):
+        bounds.append(bounds[-1] + base + (1 if i < rem else 0))
+    plan = PartitionPlan(stage_ranks=list(range(world_size)),
+                         ranges=[(bounds[i], bounds[i + 1]) for i in range(world_size)])
+    engine = PipelineEngine(comm, layer_cfgs, plan, loss_fn=torch.nn.CrossEntropyLoss(),
+                            stage_kwargs=dict(record_forward_time=False))
+    start, end = plan.ranges[engine.stage_idx]
+    engine.stage.load_layer_state_dicts(
+        [
+            {k: v.detach().clone() for k, v in full.module[i].state_dict().items()}
+            for i in range(start, end)
+        ]
+    )
+    opt = FusedSGD(engine.parameters(), lr=lr)
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        losses.append(engine.run_iteration(batch, labels, num_microbatches=M,
+                                           schedule="1f1b"))
+        opt.step()
+    if rank == 0:
+        np.save(f"{out_dir}/losses_1f1b4.npy", np.array(losses, dtype=np.float64))
+    comm.barrier()
+    from skycomputing_amd.parallel import destroy
+
+    destroy()
+
+
+def test_four_stage_1f1b_matches_local(tmp_path):
+    """1F1B at 4 stages with M=8 (deeper warmup/steady/cooldown phases
+    than the 3-stage case) — losses must match the local reference."""
+    layer_cfgs = tiny_bert_cfg(2)
+    batch, labels = _make_batch()
+    lr, steps, M = 0.05, 2, 8
+    ref = _single_process_reference(layer_cfgs, batch, labels, lr, steps, M)
+    run_multiprocess(_four_stage_1f1b_worker, 4, 29920, layer_cfgs, batch,
+                     labels, lr, steps, M, str(tmp_path))
+    got = np.load(f"{tmp_path}/losses_1f1b4.npy")
+    assert np.allclose(got, np.array(ref), rtol=1e-4, atol=1e-5), (got, ref)
