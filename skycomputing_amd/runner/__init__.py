@@ -1,7 +1,8 @@
 from .hooks import Hook
 from .hooks_collection import (
-    CheckpointHook, LRScheduleHook, MetricsHook, StopHook, TimerHook,
+    CheckpointHook, LRScheduleHook, MemoryHook, MetricsHook, StopHook,
+    TimerHook,
 )
 from .runner import Runner
 
-__all__ = ["Runner", "Hook", "CheckpointHook", "LRScheduleHook", "MetricsHook", "StopHook", "TimerHook"]
+__all__ = ["Runner", "Hook", "CheckpointHook", "LRScheduleHook", "MemoryHook", "MetricsHook", "StopHook", "TimerHook"]
