@@ -5,9 +5,10 @@ A from-scratch rebuild of the capabilities of hpcaitech/SkyComputing
 
   * process-per-GPU SPMD over RCCL/xGMI instead of host-orchestrated
     TensorPipe RPC;
-  * hand-written HIP/CDNA4 kernels for the BERT hot path (fused LayerNorm,
-    bias-GELU, masked softmax, fused embedding, multi-tensor SGD; MFMA GEMM
-    fusions) instead of eager CUDA ops + optional apex;
+  * hand-written HIP/CDNA4 kernels for the BERT hot path (fully-fused
+    MFMA attention fwd+bwd, fused LayerNorm(+residual+dropout), bias-GELU,
+    masked softmax, fused embedding, multi-tensor SGD, MFMA GEMM with
+    fused epilogues) instead of eager CUDA ops + optional apex;
   * benchmark-driven even/dynamic/optimal layer allocation with an EXACT
     subset-DP solver replacing the reference's 300s/20%-gap MIP;
   * device-resident activations, bf16 compute, 288 GB HBM3E sizing.
