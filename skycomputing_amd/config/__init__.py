@@ -9,9 +9,9 @@ child's keys overriding the base's.
 
 from __future__ import annotations
 
+import importlib.util
 import os
 import types
-from importlib.machinery import SourceFileLoader
 
 
 class Config(dict):
@@ -57,7 +57,9 @@ def _py2dict(path: str) -> dict:
         raise FileNotFoundError(path)
     if not path.endswith(".py"):
         raise ValueError(f"config file must be a .py file, got {path}")
-    module = SourceFileLoader("_sky_config", path).load_module()
+    spec = importlib.util.spec_from_file_location("_sky_config", path)
+    module = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(module)
     cfg = {
         k: v
         for k, v in vars(module).items()
