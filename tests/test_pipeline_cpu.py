@@ -48,7 +48,7 @@ def _single_process_reference(layer_cfgs, batch, labels, lr, steps, num_microbat
                 out = stage(*(x[m] for x in mb_in))
                 loss = loss_fn(out, mb_lab[m])
                 (loss / M).backward()
-                tot += float(loss)
+                tot += float(loss.detach())
             losses.append(tot / M)
         opt.step()
     return losses
