@@ -25,7 +25,7 @@ def tiny_bert_cfg(num_encoder_layers: int = 2, dropout: float = 0.0) -> list[dic
 
 
 def run_multiprocess(fn, world_size: int, port: int, *args, timeout: float = 180.0,
-                     retries: int = 1):
+                     retries: int = 2):
     """Spawn `world_size` ranks running fn(rank, world_size, *args); raise on
     failure. The rendezvous port is jittered by pid to avoid TIME_WAIT
     collisions; transient rendezvous/teardown failures are retried once on

@@ -40,7 +40,7 @@ def test_bench_json_contract(tmp_path):
     assert json.loads(out.read_text()) == d
 
 
-def _torchrun_bench(nproc, extra, out, attempts=2):
+def _torchrun_bench(nproc, extra, out, attempts=3):
     """Run bench.py under torch.distributed.run; one retry on a fresh
     rendezvous port (loaded machines occasionally drop the first
     rendezvous or overrun the startup window)."""
