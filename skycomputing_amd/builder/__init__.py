@@ -76,4 +76,12 @@ def build_dataloader_from_cfg(batch_size: int, dataset_cfg: dict, **loader_kwarg
     dataset = build_dataset(dataset_cfg)
     defaults = dict(shuffle=False, drop_last=True, num_workers=0)
     defaults.update(loader_kwargs)
+    # Every rank iterates its own DataLoader and the pipeline pairs the
+    # first stage's inputs with the last stage's labels, so iteration MUST
+    # be rank-identical. With shuffle=True that only holds if the RNG is
+    # shared: pin a fixed-seed generator unless the caller supplied one.
+    if defaults.get("shuffle") and "generator" not in defaults:
+        g = torch.Generator()
+        g.manual_seed(0x5C1C)
+        defaults["generator"] = g
     return torch.utils.data.DataLoader(dataset, batch_size=batch_size, **defaults)

@@ -163,8 +163,15 @@ class StageModule(nn.Module):
 
     def get_layer_state_dicts(self) -> list[dict]:
         """Per-layer CPU state dicts in stage order (checkpoint contract)."""
+        # floating tensors are normalized to fp32 (partition/dtype-portable
+        # contract); integer buffers (e.g. BatchNorm num_batches_tracked)
+        # keep their dtype — casting them through float loses exactness
         return [
-            {k: v.detach().to("cpu", torch.float32) for k, v in layer.state_dict().items()}
+            {
+                k: (v.detach().to("cpu", torch.float32)
+                    if v.is_floating_point() else v.detach().cpu())
+                for k, v in layer.state_dict().items()
+            }
             for layer in self.module
         ]
 

@@ -143,13 +143,18 @@ def sgd_step(
         if g is None:
             continue
         target = master_params[i] if master_params is not None else p
-        gf = g.to(target.dtype)
+        # the update is computed in fp32 (momentum buffers are fp32 even
+        # when the params are bf16 without masters — see FusedSGD)
+        gf = g.float()
         if weight_decay != 0.0:
-            gf = gf.add(target, alpha=weight_decay)
+            gf = gf.add(target.float(), alpha=weight_decay)
         if momentum != 0.0:
             buf = momentum_bufs[i]
             buf.mul_(momentum).add_(gf)
             gf = buf
-        target.add_(gf, alpha=-lr)
+        if target.dtype == torch.float32:
+            target.add_(gf, alpha=-lr)
+        else:
+            target.copy_((target.float().add_(gf, alpha=-lr)).to(target.dtype))
         if master_params is not None:
             p.copy_(target.to(p.dtype))

@@ -36,13 +36,31 @@ class FusedSGD:
         self.masters = (
             [p.detach().clone().float() for p in self.params] if master_weights else None
         )
+        # momentum buffers are ALWAYS fp32: the HIP multi-tensor path reads
+        # them as float* (ops/hip/sgd.hip) regardless of the param dtype,
+        # and fp32 accumulation is the numerically right choice for bf16
+        # params even without master weights.
         self.momentum_bufs = (
-            [torch.zeros_like(self.masters[i] if self.masters else p)
-             for i, p in enumerate(self.params)]
+            [torch.zeros(p.shape, dtype=torch.float32, device=p.device)
+             for p in self.params]
             if momentum != 0.0
             else None
         )
         self._gpu_plan = None  # lazy multi-tensor launch plan
+
+    @torch.no_grad()
+    def sync_masters(self):
+        """Refresh fp32 masters (and reset momentum) from the CURRENT param
+        values. Must be called after any out-of-band weight mutation
+        (checkpoint restore, ParameterServer scatter) or the next step()
+        would write ``stale_master - lr*grad`` back over the restored
+        weights."""
+        if self.masters is not None:
+            for m, p in zip(self.masters, self.params):
+                m.copy_(p.detach().float())
+        if self.momentum_bufs is not None:
+            for b in self.momentum_bufs:
+                b.zero_()
 
     def zero_grad(self, set_to_none: bool = True):
         for p in self.params:

@@ -57,6 +57,12 @@ class CheckpointHook(Hook):
         if runner.comm.rank == 0:
             meta = ps.load_weights_from_file(self.load_from)
         ps.scatter_to_engine(runner.engine, runner.comm)
+        # the optimizer cloned its fp32 masters at construction time, i.e.
+        # BEFORE this restore — refresh them (and reset momentum) or the
+        # first step() writes stale_master - lr*grad over the restored
+        # weights
+        if hasattr(runner.optimizer, "sync_masters"):
+            runner.optimizer.sync_masters()
         if self.resume_counters:
             meta = runner.comm.broadcast_object(meta, src=0) or {}
             runner.epoch = int(meta.get("epoch", 0))

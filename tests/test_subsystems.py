@@ -139,6 +139,33 @@ def test_fused_sgd_state_dict_roundtrip():
     assert torch.allclose(opt2.momentum_bufs[0], opt.momentum_bufs[0])
 
 
+def test_fused_sgd_sync_masters_after_restore():
+    # out-of-band weight restore + stale fp32 masters
+    # must not revert params on the next step. lr=0 step must be a no-op
+    # after sync_masters().
+    p = [torch.randn(8, dtype=torch.bfloat16, requires_grad=True)]
+    opt = FusedSGD(p, lr=0.0, momentum=0.9)  # masters auto-enabled for bf16
+    assert opt.masters is not None
+    restored = torch.randn(8, dtype=torch.bfloat16)
+    with torch.no_grad():
+        p[0].copy_(restored)  # simulates checkpoint restore
+    opt.sync_masters()
+    p[0].grad = torch.randn(8, dtype=torch.bfloat16)
+    opt.step()
+    assert torch.equal(p[0].detach(), restored)
+
+
+def test_fused_sgd_bf16_momentum_bufs_are_fp32():
+    # momentum buffers must be fp32 regardless of
+    # param dtype (the HIP multi-tensor kernel reads them as float*).
+    p = [torch.randn(8, dtype=torch.bfloat16, requires_grad=True)]
+    opt = FusedSGD(p, lr=0.1, momentum=0.9, master_weights=False)
+    assert opt.momentum_bufs[0].dtype == torch.float32
+    p[0].grad = torch.randn(8, dtype=torch.bfloat16)
+    opt.step()  # eager path must handle fp32 bufs + bf16 params
+    assert torch.isfinite(p[0].float()).all()
+
+
 def test_parameter_server_save_load(tmp_path):
     ps = ParameterServer(2)
     ps.update_weights({"w": torch.randn(3)}, 0)
