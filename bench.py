@@ -40,6 +40,8 @@ def parse_args():
     p.add_argument("--slowdowns", default="")
     p.add_argument("--stimulate", action="store_true")
     p.add_argument("--dropout", type=float, default=0.1)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"],
+                   help="compute dtype on GPU (CPU runs are always fp32)")
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph step capture (1-GPU path)")
     p.add_argument("--json-out", default="")
@@ -61,7 +63,9 @@ def main():
     comm = init_distributed()
     world = comm.world_size
     rank = comm.rank
-    dtype = torch.bfloat16 if use_cuda else torch.float32
+    dtype = (
+        torch.bfloat16 if (use_cuda and args.dtype == "bf16") else torch.float32
+    )
     torch.manual_seed(1234 + rank)
 
     M = args.microbatches or (1 if world == 1 else 8)
@@ -216,7 +220,7 @@ def main():
             "higher_is_better": False,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": "bf16" if use_cuda else "fp32",
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
             "data": "synthetic",
             "config": {
                 "model": f"bert-{args.layers}L-H1024-A16",
