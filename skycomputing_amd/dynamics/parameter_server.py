@@ -76,16 +76,39 @@ class ParameterServer:
                         self._layers[start + off] = sd
 
     def scatter_to_engine(self, engine, comm):
-        """Collective: rank 0 broadcasts the full layer list; each rank
-        loads its slice. (Layer state dicts are small relative to HBM; a
-        per-rank sharded variant is the natural next step.)"""
-        layers = comm.broadcast_object(self._layers if comm.rank == 0 else None, src=0)
+        """Collective: rank 0 sends each rank ONLY its slice(s) of the
+        layer list (point-to-point), instead of broadcasting the full
+        model to every rank — at 160L/8 ranks a full fp32 broadcast is
+        ~8 GB per rank; the sharded transfer is ~1/world of that. The
+        partition plan is known on every rank, so the send/recv schedule
+        is deterministic without a handshake."""
+        plan = engine.plan
+        # per-rank list of (start, end) slices, in stage order (a rank owns
+        # several slices under the interleaved engine's virtual stages)
+        rank_slices: dict[int, list[tuple[int, int]]] = {}
+        for s, rng in enumerate(plan.ranges):
+            rank_slices.setdefault(plan.stage_ranks[s], []).append(tuple(rng))
+
+        my_payload = None
         if comm.rank == 0:
-            self._layers = layers
+            for r in sorted(rank_slices):
+                payload = [
+                    (start, self._layers[start:end])
+                    for start, end in rank_slices[r]
+                ]
+                if r == 0:
+                    my_payload = payload
+                else:
+                    comm.send_object(payload, dst=r)
+        elif comm.rank in rank_slices:
+            my_payload = comm.recv_object(src=0)
+        if my_payload is None:
+            return
+        by_start = {start: dicts for start, dicts in my_payload}
         if hasattr(engine, "chunks"):  # InterleavedPipelineEngine
             for s, chunk in engine.chunks.items():
-                start, end = engine.plan.ranges[s]
-                chunk.load_layer_state_dicts(layers[start:end])
+                start, _end = engine.plan.ranges[s]
+                chunk.load_layer_state_dicts(by_start[start])
         elif engine.stage_idx is not None:
-            start, end = engine.plan.ranges[engine.stage_idx]
-            engine.stage.load_layer_state_dicts(layers[start:end])
+            start, _end = engine.plan.ranges[engine.stage_idx]
+            engine.stage.load_layer_state_dicts(by_start[start])
