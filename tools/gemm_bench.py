@@ -38,6 +38,72 @@ def sky_gemm(a, b, bias=None, transA=0, transB=1, epi=0, z=None):
     return c
 
 
+_WK = {}
+
+
+def sky_gemm2(a, b, bias=None, epi=0, z=None, gsu=1):
+    """v2 256^2 8-phase NT forward: C = a @ b.T (+bias)(+gelu)."""
+    lib = hiplib.require()
+    M, K = a.shape
+    N = b.shape[0]
+    c = torch.empty(M, N, dtype=a.dtype, device=a.device)
+    wk = None
+    if gsu > 1:
+        key = (M, N, gsu)
+        wk = _WK.get(key)
+        if wk is None:
+            wk = torch.empty(gsu * M * N, dtype=torch.float32, device=a.device)
+            _WK[key] = wk
+    check(
+        lib.sky_gemm2(
+            torch.cuda.current_stream().cuda_stream, ptr(a), ptr(b), ptr(c),
+            ptr(bias), ptr(z), ptr(wk), M, N, K, a.stride(0), b.stride(0),
+            c.stride(0), epi, gsu,
+        ),
+        "sky_gemm2",
+    )
+    return c
+
+
+def bench_v2():
+    print("\n-- v2 256^2 8-phase (NT fwd) --")
+    print(f"{'shape':<18} {'gsu':>3} {'torch us':>9} {'v2 us':>9} {'ratio':>6} {'v2 TF':>8}  max_err")
+    shapes = [
+        ("qkv 4096x3072", 4096, 3072, 1024, (1, 2)),
+        ("proj 4096x1024", 4096, 1024, 1024, (1, 2, 4)),
+        ("ffnup 4096x4096", 4096, 4096, 1024, (1, 2)),
+        ("ffndn 4096x1024", 4096, 1024, 4096, (1, 2, 4, 8)),
+        ("sq 4096x4096x4096", 4096, 4096, 4096, (1,)),
+    ]
+    for name, M, N, K, gsus in shapes:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.03
+        ref = x.float() @ w.float().t()
+        tt = bench(lambda: torch.nn.functional.linear(x, w))
+        for gsu in gsus:
+            c = sky_gemm2(x, w, gsu=gsu)
+            err = (c.float() - ref).abs().max().item()
+            rel = err / ref.abs().max().item()
+            ts = bench(lambda: sky_gemm2(x, w, gsu=gsu))
+            tf = 2 * M * N * K / ts / 1e12
+            print(f"{name:<18} {gsu:>3} {tt*1e6:9.1f} {ts*1e6:9.1f} {tt/ts:6.2f} {tf:8.0f}  {err:.3f} ({rel:.2e})")
+    # fused bias+gelu epilogue + Z store
+    M, N, K = 4096, 4096, 1024
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.03
+    b = torch.randn(N, dtype=torch.bfloat16, device="cuda")
+    z = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    import math
+    zr = x.float() @ w.float().t() + b.float()
+    ref = zr * 0.5 * (1 + torch.erf(zr / math.sqrt(2)))
+    c = sky_gemm2(x, w, bias=b, epi=2, z=z)
+    err = (c.float() - ref).abs().max().item()
+    errz = (z.float() - zr).abs().max().item()
+    ts = bench(lambda: sky_gemm2(x, w, bias=b, epi=2, z=z))
+    print(f"{'ffnup+gelu(+z)':<18} {1:>3} {'':>9} {ts*1e6:9.1f} {'':>6} "
+          f"{2*M*N*K/ts/1e12:8.0f}  {err:.3f}/{errz:.3f}")
+
+
 def bench(fn, iters=30):
     for _ in range(5):
         fn()
@@ -112,4 +178,9 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    if "--v2-only" in sys.argv:
+        torch.manual_seed(0)
+        bench_v2()
+    else:
+        main()
+        bench_v2()
