@@ -186,7 +186,10 @@ class GlueDataset(Dataset):
         rows = []
         with open(path, encoding="utf-8") as f:
             reader = csv.reader(f, delimiter="\t", quotechar=None)
-            header = next(reader)
+            # CoLA tsvs have NO header row (reference ColaProcessor reads
+            # from line 0, scaelum/dataset/glue/processor.py); the other
+            # tasks carry one.
+            header = next(reader) if task != "cola" else None
             if task == "mnli":
                 idx = (header.index("sentence1"), header.index("sentence2"), header.index("gold_label"))
                 for r in reader:

@@ -100,8 +100,11 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
   const int lm = l & 15, lg = l >> 4;
   const int wr = w >> 2, wc = w & 3;  // 8 waves as 2M x 4N
 
-  // XCD-aware bijective remap (guide T1), then tile-major decode so
-  // consecutive blocks on one XCD share A rows (same g for split-K).
+  // XCD-aware bijective remap (guide T1), then a supertile decode: within
+  // an XCD's contiguous logical range, tiles cover a 4(M)x*(N) rectangle
+  // instead of a full M-row — the 32-tile group's A panels (2 MB) AND a
+  // few B panels co-reside in the XCD's 4 MB L2, cutting HBM re-reads
+  // (row-major order re-reads every B panel per XCD).
   int bid = blockIdx.x;
   {
     const int nwg = gridDim.x;
@@ -110,9 +113,16 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
     bid = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
   }
   const int ntn = N / G2_BN;
-  const int ntiles = (M / G2_BM) * ntn;
+  const int ntm = M / G2_BM;
+  const int ntiles = ntm * ntn;
   const int g = SPLIT ? bid / ntiles : 0;
-  const int tile = SPLIT ? bid % ntiles : bid;
+  int tile = SPLIT ? bid % ntiles : bid;
+  {
+    const int SM = (ntm >= 4) ? 4 : ntm;  // supertile height (m tiles)
+    const int stripe = SM * ntn;
+    const int mb = tile / stripe, r = tile % stripe;
+    tile = (mb * SM + r % SM) * ntn + r / SM;
+  }
   const int m0 = (tile / ntn) * G2_BM;
   const int n0 = (tile % ntn) * G2_BN;
   const int kpg = SPLIT ? K / gsu : K;
@@ -185,17 +195,17 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
   for (int t = 0; t < NT; ++t) {
     char* abuf = lds + (t & 1) * 65536;
     char* bbuf = abuf + 32768;
-    // phase 0: quadrant (0,0); issue A1[t+1], B0[t+1]
+    // phase 0: quadrant (0,0); issue A1[t+1]
     readA(abuf, 0);
     readB(bbuf, 0);
     issueA(t + 1, 1);
-    issueB(t + 1, 0);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     mma(0, 0);
     __builtin_amdgcn_s_barrier();
-    // phase 1: quadrant (0,1); af reused
+    // phase 1: quadrant (0,1); af reused; issue B0[t+1]
     readB(bbuf, 1);
+    issueB(t + 1, 0);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     mma(0, 1);
