@@ -56,6 +56,21 @@ DEV int g2_swz(int byte_addr, int row) { return byte_addr ^ ((row & 7) << 4); }
 
 enum { G2_EPI_NONE = 0, G2_EPI_BIAS = 1, G2_EPI_BIAS_GELU = 2 };
 
+// operand storage modes: DIRECT = stored [out][red] (reduction contiguous,
+// staged as [out][64k] images read by ds_read_b128); KMAJOR = stored
+// [red][out] (out contiguous — dgrad's W, wgrad's dY/X), staged as
+// [half][64k][128out] images (glds along out) and read TRANSPOSED with
+// ds_read_b64_tr_b16 (semantics measured by tools/tr16_probe.hip:
+// result(lane 4q+r, reg j) = elem at lane(4j+q).addr + r).
+enum { G2_DIRECT = 0, G2_KMAJOR = 1 };
+
+typedef __attribute__((ext_vector_type(4))) short s16x4;
+typedef __attribute__((address_space(3))) s16x4* g2_las4;
+
+// k-dependent XOR on within-half column quads (bank spread for the tr16
+// reads; bits 2-4 of the quad index, preserves 16-B staging granularity)
+DEV int g2_kswz(int k) { return 4 * ((((k >> 3) & 1) << 2) | (k & 3)); }
+
 // ---- half-tile staging (16 KB, 2 x glds16 per thread) ----
 // A image: [256][64] bf16 row-major at abuf; half h = rows {h*64..h*64+63,
 // 128+h*64..}; chunk c covers rows c*128+h*64+rl.
@@ -87,7 +102,28 @@ DEV void g2_stageB(const ushort_t* __restrict__ B, int ldb, int n0, int k0,
   }
 }
 
-template <int EPI, bool STORE_Z, bool SPLIT>
+// KMAJOR half staging: operand stored [red][out] (ld = out-row stride);
+// image = [half 16KB][k 0..63][16 x 16B slots]; slot v16 of row k holds the
+// 8 source columns starting at out-offset of quad Q = (2*v16) ^ g2_kswz(k).
+// outsel(h, Q) maps a within-half quad to the tile-local out coordinate:
+//   A halves split on out bit6: out = (Q>>4)*128 + h*64 + (Q&15)*4
+//   B halves split on out bit5: out = (Q>>3)*64  + h*32 + (Q&7)*4
+template <bool BHALF>
+DEV void g2_stage_kmaj(const ushort_t* __restrict__ X, int ld, int out0,
+                       int k0, char* img, int h, int tid) {
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    const int u = tid + c * 512;
+    const int k = u >> 4, v16 = u & 15;
+    const int Q = (2 * v16) ^ g2_kswz(k);
+    const int out = BHALF ? (Q >> 3) * 64 + h * 32 + (Q & 7) * 4
+                          : (Q >> 4) * 128 + h * 64 + (Q & 15) * 4;
+    const ushort_t* g = X + (size_t)(k0 + k) * ld + out0 + out;
+    g2_glds16(g, img + h * 16384 + k * 256 + v16 * 16);
+  }
+}
+
+template <int AM, int BM, int EPI, bool STORE_Z, bool SPLIT>
 __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
     ushort_t* __restrict__ C, const ushort_t* __restrict__ bias,
@@ -132,12 +168,18 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
   auto issueA = [&](int t, int h) {
     char* abuf = lds + (t & 1) * 65536;
     const int tt = t < NT ? t : NT - 1;  // clamp SOURCE only; slot stays t&1
-    g2_stageA(A, lda, m0, kbase + tt * G2_BK, abuf, h, tid);
+    if (AM == G2_DIRECT)
+      g2_stageA(A, lda, m0, kbase + tt * G2_BK, abuf, h, tid);
+    else
+      g2_stage_kmaj<false>(A, lda, m0, kbase + tt * G2_BK, abuf, h, tid);
   };
   auto issueB = [&](int t, int h) {
     char* bbuf = lds + (t & 1) * 65536 + 32768;
     const int tt = t < NT ? t : NT - 1;
-    g2_stageB(B, ldb, n0, kbase + tt * G2_BK, bbuf, h, tid);
+    if (BM == G2_DIRECT)
+      g2_stageB(B, ldb, n0, kbase + tt * G2_BK, bbuf, h, tid);
+    else
+      g2_stage_kmaj<true>(B, ldb, n0, kbase + tt * G2_BK, bbuf, h, tid);
   };
 
   f32x4 acc[8][4];
@@ -159,24 +201,52 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
   bf16x8 af[4][2], bfr[2][2];
   const int swz_x = (lm & 7) << 4;  // row&7 == lm&7 for every fragment row
 
+  const int s4 = l & 15;  // tr16 source-lane index within the 16-lane group
+
+  // transposed fragment read from a KMAJOR image: one tr16 pair yields the
+  // 8 k-values of out-row (frag base + lm); lane s supplies the address of
+  // (k-row k0 + s>>2, within-half quad bq + (s&3)) per the probed mapping.
+  auto read_kmaj = [&](char* img, int h, int bq, int ks) -> bf16x8 {
+    s16x4 v[2];
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int kl = ks * 32 + lg * 8 + half * 4 + (s4 >> 2);
+      const int byte = h * 16384 + kl * 256 +
+                       (((bq + (s4 & 3)) ^ g2_kswz(kl)) << 3);
+      v[half] = __builtin_amdgcn_ds_read_tr16_b64_v4i16((g2_las4)(img + byte));
+    }
+    return (bf16x8){v[0][0], v[0][1], v[0][2], v[0][3],
+                    v[1][0], v[1][1], v[1][2], v[1][3]};
+  };
+
   auto readA = [&](char* abuf, int mh) {
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const int row = wr * 128 + mh * 64 + i * 16 + lm;
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
-        af[i][ks] = *(const bf16x8*)(abuf +
-                                     ((row * 128 + (ks * 32 + lg * 8) * 2) ^ swz_x));
+      for (int ks = 0; ks < 2; ++ks) {
+        if (AM == G2_DIRECT) {
+          const int row = wr * 128 + mh * 64 + i * 16 + lm;
+          af[i][ks] = *(const bf16x8*)(abuf +
+                                       ((row * 128 + (ks * 32 + lg * 8) * 2) ^ swz_x));
+        } else {
+          af[i][ks] = read_kmaj(abuf, mh, wr * 16 + i * 4, ks);
+        }
+      }
     }
   };
   auto readB = [&](char* bbuf, int nh) {
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      const int n = wc * 64 + nh * 32 + j * 16 + lm;
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
-        bfr[j][ks] = *(const bf16x8*)(bbuf +
-                                      ((n * 128 + (ks * 32 + lg * 8) * 2) ^ swz_x));
+      for (int ks = 0; ks < 2; ++ks) {
+        if (BM == G2_DIRECT) {
+          const int n = wc * 64 + nh * 32 + j * 16 + lm;
+          bfr[j][ks] = *(const bf16x8*)(bbuf +
+                                        ((n * 128 + (ks * 32 + lg * 8) * 2) ^ swz_x));
+        } else {
+          bfr[j][ks] = read_kmaj(bbuf, nh, wc * 8 + j * 4, ks);
+        }
+      }
     }
   };
   auto mma = [&](int mh, int nh) {
@@ -319,7 +389,8 @@ static int g2_set_lds(const void* f) {
 SKY_EXPORT int sky_gemm2(uint64_t stream, uint64_t A, uint64_t B, uint64_t C,
                          uint64_t bias, uint64_t Z, uint64_t Wk, int64_t M,
                          int64_t N, int64_t K, int64_t lda, int64_t ldb,
-                         int64_t ldc, int epi, int gsu) {
+                         int64_t ldc, int transA, int transB, int epi,
+                         int gsu) {
   if (M % G2_BM || N % G2_BN || K % G2_BK) return (int)hipErrorInvalidValue;
   if (gsu > 1 && (K % (gsu * G2_BK) || !Wk)) return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
@@ -328,27 +399,42 @@ SKY_EXPORT int sky_gemm2(uint64_t stream, uint64_t A, uint64_t B, uint64_t C,
   const bool split = gsu > 1;
   const bool sz = Z != 0;
   int rc = 0;
-#define G2K(EP, SZ, SP)                                                        \
+#define G2K(AM, BM, EP, SZ, SP)                                                \
   do {                                                                         \
-    rc = g2_set_lds((const void*)&gemm2_kernel<EP, SZ, SP>);                   \
+    rc = g2_set_lds((const void*)&gemm2_kernel<AM, BM, EP, SZ, SP>);           \
     if (rc) return rc;                                                         \
-    hipLaunchKernelGGL((gemm2_kernel<EP, SZ, SP>), grid, dim3(G2_BLOCK),       \
-                       G2_LDS, s, (const ushort_t*)A, (const ushort_t*)B,      \
-                       (ushort_t*)C, (const ushort_t*)bias, (ushort_t*)Z,      \
-                       (float*)Wk, (int)M, (int)N, (int)K, (int)lda, (int)ldb, \
-                       (int)ldc, gsu);                                         \
+    hipLaunchKernelGGL((gemm2_kernel<AM, BM, EP, SZ, SP>), grid,               \
+                       dim3(G2_BLOCK), G2_LDS, s, (const ushort_t*)A,          \
+                       (const ushort_t*)B, (ushort_t*)C,                       \
+                       (const ushort_t*)bias, (ushort_t*)Z, (float*)Wk,        \
+                       (int)M, (int)N, (int)K, (int)lda, (int)ldb, (int)ldc,   \
+                       gsu);                                                   \
   } while (0)
-  // split kernels ignore the epilogue (applied in the reduce)
-  if (split) {
-    G2K(G2_EPI_NONE, false, true);
-  } else if (epi == G2_EPI_NONE) {
-    G2K(G2_EPI_NONE, false, false);
-  } else if (epi == G2_EPI_BIAS) {
-    G2K(G2_EPI_BIAS, false, false);
-  } else if (sz) {
-    G2K(G2_EPI_BIAS_GELU, true, false);
+  // supported orientations: NT fwd (direct,direct) with full epilogues;
+  // dgrad NN (direct,kmajor) and wgrad TN (kmajor,kmajor) epilogue-free
+  if (transA == 0 && transB == 0) {
+    // split kernels ignore the epilogue (applied in the reduce)
+    if (split) {
+      G2K(G2_DIRECT, G2_DIRECT, G2_EPI_NONE, false, true);
+    } else if (epi == G2_EPI_NONE) {
+      G2K(G2_DIRECT, G2_DIRECT, G2_EPI_NONE, false, false);
+    } else if (epi == G2_EPI_BIAS) {
+      G2K(G2_DIRECT, G2_DIRECT, G2_EPI_BIAS, false, false);
+    } else if (sz) {
+      G2K(G2_DIRECT, G2_DIRECT, G2_EPI_BIAS_GELU, true, false);
+    } else {
+      G2K(G2_DIRECT, G2_DIRECT, G2_EPI_BIAS_GELU, false, false);
+    }
+  } else if (transA == 0 && transB == 1) {
+    if (epi != G2_EPI_NONE && !split) return (int)hipErrorInvalidValue;
+    if (split) G2K(G2_DIRECT, G2_KMAJOR, G2_EPI_NONE, false, true);
+    else G2K(G2_DIRECT, G2_KMAJOR, G2_EPI_NONE, false, false);
+  } else if (transA == 1 && transB == 1) {
+    if (epi != G2_EPI_NONE && !split) return (int)hipErrorInvalidValue;
+    if (split) G2K(G2_KMAJOR, G2_KMAJOR, G2_EPI_NONE, false, true);
+    else G2K(G2_KMAJOR, G2_KMAJOR, G2_EPI_NONE, false, false);
   } else {
-    G2K(G2_EPI_BIAS_GELU, false, false);
+    return (int)hipErrorInvalidValue;
   }
 #undef G2K
   LAUNCH_CHECK();

@@ -61,8 +61,9 @@ def _register_signatures(lib):
         "sky_glds_probe": [u64, u64, u64, i32],
         "sky_gemm": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, i64, i64, i32, i32, i32, i32, i32],
         # strm A B C bias Z | M N K lda ldb ldc | transA transB epi dt use_glds
-        "sky_gemm2": [u64, u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, i64, i64, i32, i32],
-        # strm A B C bias Z Wk | M N K lda ldb ldc | epi gsu  (NT fwd, 256^2 8-phase)
+        "sky_gemm2": [u64, u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, i64, i64, i32, i32, i32, i32],
+        # strm A B C bias Z Wk | M N K lda ldb ldc | transA transB epi gsu
+        # (256^2 8-phase; transA/B: 0=direct [out][red], 1=kmajor [red][out])
         "sky_pack3": [u64, u64, u64, u64, u64, i64, i64, i64, i64, i32],
         "sky_attn_fwd": [u64, u64, u64, u64, u64, u64, i64, i64, i64, i64, f32, f32, u64, u64],
         #                strm qkv  mask out  m    l    B    S    h    d  scale keep salt state

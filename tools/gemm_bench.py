@@ -41,11 +41,15 @@ def sky_gemm(a, b, bias=None, transA=0, transB=1, epi=0, z=None):
 _WK = {}
 
 
-def sky_gemm2(a, b, bias=None, epi=0, z=None, gsu=1):
-    """v2 256^2 8-phase NT forward: C = a @ b.T (+bias)(+gelu)."""
+def sky_gemm2(a, b, bias=None, epi=0, z=None, gsu=1, transA=0, transB=0):
+    """v2 256^2 8-phase GEMM. transA/transB: 0 = operand stored [out][red],
+    1 = stored [red][out] (kmajor; read via ds_read_b64_tr_b16).
+    NT fwd: C=a@b.T (tA=0,tB=0); dgrad NN: C=a@b (tA=0,tB=1);
+    wgrad TN: C=a.T@b (tA=1,tB=1)."""
     lib = hiplib.require()
-    M, K = a.shape
-    N = b.shape[0]
+    M = a.shape[1] if transA else a.shape[0]
+    K = a.shape[0] if transA else a.shape[1]
+    N = b.shape[1] if transB else b.shape[0]
     c = torch.empty(M, N, dtype=a.dtype, device=a.device)
     wk = None
     if gsu > 1:
@@ -58,7 +62,7 @@ def sky_gemm2(a, b, bias=None, epi=0, z=None, gsu=1):
         lib.sky_gemm2(
             torch.cuda.current_stream().cuda_stream, ptr(a), ptr(b), ptr(c),
             ptr(bias), ptr(z), ptr(wk), M, N, K, a.stride(0), b.stride(0),
-            c.stride(0), epi, gsu,
+            c.stride(0), transA, transB, epi, gsu,
         ),
         "sky_gemm2",
     )
@@ -87,6 +91,27 @@ def bench_v2():
             ts = bench(lambda: sky_gemm2(x, w, gsu=gsu))
             tf = 2 * M * N * K / ts / 1e12
             print(f"{name:<18} {gsu:>3} {tt*1e6:9.1f} {ts*1e6:9.1f} {tt/ts:6.2f} {tf:8.0f}  {err:.3f} ({rel:.2e})")
+    # transposed orientations (dgrad NN, wgrad TN) via tr16 kmajor reads
+    print("  -- transposed (kmajor) --")
+    for name, M, N, K, tA, tB, gsus in [
+        ("dgradNN 4096x1024", 4096, 1024, 4096, 0, 1, (1, 2, 4)),
+        ("dgradNN 4096x4096", 4096, 4096, 1024, 0, 1, (1,)),
+        ("wgradTN 4096x1024", 4096, 1024, 4096, 1, 1, (1, 2, 4)),
+        ("wgradTN 1024x4096", 1024, 4096, 4096, 1, 1, (1, 2, 4)),
+    ]:
+        a = (torch.randn(K, M, dtype=torch.bfloat16, device="cuda") * 0.05
+             if tA else torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.05)
+        b = torch.randn(K, N, dtype=torch.bfloat16, device="cuda") * 0.03
+        ref = (a.float().t() if tA else a.float()) @ b.float()
+        tt = bench(lambda: (a.t() if tA else a) @ b)
+        for gsu in gsus:
+            c = sky_gemm2(a, b, gsu=gsu, transA=tA, transB=tB)
+            err = (c.float() - ref).abs().max().item()
+            rel = err / ref.abs().max().item()
+            ts = bench(lambda: sky_gemm2(a, b, gsu=gsu, transA=tA, transB=tB))
+            tf = 2 * M * N * K / ts / 1e12
+            print(f"{name:<18} {gsu:>3} {tt*1e6:9.1f} {ts*1e6:9.1f} {tt/ts:6.2f} {tf:8.0f}  {err:.3f} ({rel:.2e})")
+
     # fused bias+gelu epilogue + Z store
     M, N, K = 4096, 4096, 1024
     x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
