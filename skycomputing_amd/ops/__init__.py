@@ -167,6 +167,15 @@ def linear_act(x, weight, bias, act: str = "gelu"):
         if (_use_hblt() and os.environ.get("SKY_HBLT_GELU") == "1"
                 and x.dtype == torch.float32 and weight.shape[0] % 8 == 0):
             return LinearGeluFn.apply(x, weight, bias)
+        # v2 hand-GEMM path: one 256^2 8-phase kernel with the bias+gelu
+        # epilogue fused (and the pre-activation stored for backward)
+        if x.dtype == torch.bfloat16:
+            from .functions import _g2_fit, _g2_sites
+
+            rows = x.numel() // x.shape[-1]
+            if ("fwd" in _g2_sites()
+                    and _g2_fit(rows, weight.shape[0], weight.shape[1])):
+                return LinearGeluFn.apply(x, weight, bias)
         y = torch.nn.functional.linear(x, weight)
         return bias_gelu(y, bias)
     return eager.linear_act(x, weight, bias, act)

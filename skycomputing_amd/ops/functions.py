@@ -488,6 +488,115 @@ def _use_hblt() -> bool:
     return os.environ.get("SKY_NO_HBLT") != "1"
 
 
+# ---- v2 hand GEMM dispatch (ops/hip/gemm2.hip) ----
+# SKY_GEMM2: "" / "0" = off; "1" = all sites; or a comma list of
+# {fwd,dgrad,wgrad}. Only perfect-fit shapes (M,N %256; K %64) route here;
+# everything else stays on hipBLASLt.
+
+_G2_SITES: set | None = None
+_G2_WK: dict = {}
+
+
+def _g2_sites() -> set:
+    global _G2_SITES
+    if _G2_SITES is None:
+        v = os.environ.get("SKY_GEMM2", "0").strip()
+        if v in ("", "0"):
+            _G2_SITES = set()
+        elif v == "1":
+            _G2_SITES = {"fwd", "dgrad", "wgrad"}
+        else:
+            _G2_SITES = {s.strip() for s in v.split(",") if s.strip()}
+    return _G2_SITES
+
+
+def _g2_fit(M: int, N: int, K: int) -> bool:
+    return M % 256 == 0 and N % 256 == 0 and K % 64 == 0
+
+
+def _g2_gsu(M: int, N: int, K: int) -> int:
+    # split-K only where the tile grid badly underfills 256 CUs AND K is
+    # deep enough to amortize the fp32 partial round-trip (measured:
+    # K=1024 shapes run best at gsu=1 despite 64-WG grids)
+    ntiles = (M // 256) * (N // 256)
+    if ntiles < 128 and K >= 2048 and K % 256 == 0:
+        return 4
+    return 1
+
+
+def _g2_workspace(M: int, N: int, gsu: int, device) -> torch.Tensor | None:
+    if gsu <= 1:
+        return None
+    key = (M, N, gsu, device)
+    wk = _G2_WK.get(key)
+    if wk is None:
+        wk = torch.empty(gsu * M * N, dtype=torch.float32, device=device)
+        _G2_WK[key] = wk
+    return wk
+
+
+def _g2_call(a, b, c, bias, z, M, N, K, lda, ldb, ldc, tA, tB, epi, gsu):
+    lib = hiplib.require()
+    wk = _g2_workspace(M, N, gsu, a.device)
+    check(
+        lib.sky_gemm2(
+            _stream(), ptr(a), ptr(b), ptr(c), ptr(bias), ptr(z),
+            ptr(wk) if wk is not None else 0, M, N, K, lda, ldb, ldc,
+            tA, tB, epi, gsu,
+        ),
+        "sky_gemm2",
+    )
+    return c
+
+
+def gemm2_fwd(x2, weight, bias=None, gelu=False, want_z=False):
+    """NT forward y = x2 @ weight.T (+bias)(+gelu). Returns (y, z|None) or
+    None when the shape doesn't fit or the site is disabled."""
+    if "fwd" not in _g2_sites() or x2.dtype != torch.bfloat16:
+        return None
+    M, K = x2.shape
+    N = weight.shape[0]
+    if not _g2_fit(M, N, K):
+        return None
+    epi = 2 if gelu else (1 if bias is not None else 0)
+    gsu = _g2_gsu(M, N, K)
+    if gsu > 1 and epi == 2 and want_z:
+        pass  # reduce kernel handles bias+gelu+z too
+    y = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
+    z = torch.empty_like(y) if (gelu and want_z) else None
+    _g2_call(x2, weight, y, bias, z, M, N, K, x2.stride(0), weight.stride(0),
+             y.stride(0), 0, 0, epi, gsu)
+    return y, z
+
+
+def gemm2_dgrad(dy2, weight):
+    """NN dgrad dx = dy2 @ weight (weight stored [N,K] = kmajor operand)."""
+    if "dgrad" not in _g2_sites() or dy2.dtype != torch.bfloat16:
+        return None
+    M, Kred = dy2.shape
+    N = weight.shape[1]
+    if not _g2_fit(M, N, Kred):
+        return None
+    dx = torch.empty(M, N, dtype=dy2.dtype, device=dy2.device)
+    _g2_call(dy2, weight, dx, None, None, M, N, Kred, dy2.stride(0),
+             weight.stride(0), dx.stride(0), 0, 1, 0, _g2_gsu(M, N, Kred))
+    return dx
+
+
+def gemm2_wgrad(dy2, x2):
+    """TN wgrad dw = dy2.T @ x2 (both stored [Mtok, out] = kmajor)."""
+    if "wgrad" not in _g2_sites() or dy2.dtype != torch.bfloat16:
+        return None
+    Kred, M = dy2.shape
+    N = x2.shape[1]
+    if not _g2_fit(M, N, Kred):
+        return None
+    dw = torch.empty(M, N, dtype=dy2.dtype, device=dy2.device)
+    _g2_call(dy2, x2, dw, None, None, M, N, Kred, dy2.stride(0),
+             x2.stride(0), dw.stride(0), 1, 1, 0, _g2_gsu(M, N, Kred))
+    return dw
+
+
 class LinearBiasFn(torch.autograd.Function):
     """Linear + bias with a custom backward: dgrad stays a hipBLASLt GEMM;
     wgrad is a direct hipBLASLt call whose BGRADB epilogue produces dbias
@@ -499,7 +608,12 @@ class LinearBiasFn(torch.autograd.Function):
     def forward(ctx, x, weight, bias):
         xs = x.shape
         x2 = x.reshape(-1, xs[-1])
-        y = torch.nn.functional.linear(x2, weight, bias)
+        g2 = gemm2_fwd(x2 if x2.is_contiguous() else x2.contiguous(),
+                       weight, bias)
+        if g2 is not None:
+            y = g2[0]
+        else:
+            y = torch.nn.functional.linear(x2, weight, bias)
         ctx.save_for_backward(x2, weight)
         ctx.xshape = xs
         ctx.has_bias = bias is not None
@@ -511,7 +625,12 @@ class LinearBiasFn(torch.autograd.Function):
         dy2 = dy.reshape(-1, weight.shape[0])
         if not dy2.is_contiguous():
             dy2 = dy2.contiguous()
-        dx = dy2.mm(weight).view(ctx.xshape)
+        dx = gemm2_dgrad(dy2, weight)
+        dx = dx.view(ctx.xshape) if dx is not None else dy2.mm(weight).view(ctx.xshape)
+        dw2 = gemm2_wgrad(dy2, x2 if x2.is_contiguous() else x2.contiguous())
+        if dw2 is not None:
+            db = colsum(dy2, weight.dtype) if ctx.has_bias else None
+            return dx, dw2, db
         if ctx.has_bias and _use_hblt():
             lib = hiplib.require()
             M, K = x2.shape
@@ -541,20 +660,24 @@ class LinearGeluFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias):
-        lib = hiplib.require()
         xs = x.shape
         x2 = x.reshape(-1, xs[-1]).contiguous()
         M, K = x2.shape
         N = weight.shape[0]
-        y = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
-        aux = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
-        check(
-            lib.sky_hblt_linear_gelu_aux(
-                _stream(), ptr(x2), ptr(weight), ptr(bias), ptr(y), ptr(aux),
-                M, N, K, _dt(x2), _dt(aux)
-            ),
-            "sky_hblt_linear_gelu_aux",
-        )
+        g2 = gemm2_fwd(x2, weight, bias, gelu=True, want_z=True)
+        if g2 is not None:
+            y, aux = g2
+        else:
+            lib = hiplib.require()
+            y = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
+            aux = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
+            check(
+                lib.sky_hblt_linear_gelu_aux(
+                    _stream(), ptr(x2), ptr(weight), ptr(bias), ptr(y), ptr(aux),
+                    M, N, K, _dt(x2), _dt(aux)
+                ),
+                "sky_hblt_linear_gelu_aux",
+            )
         ctx.save_for_backward(x2, weight, aux)
         ctx.xshape = xs
         return y.view(*xs[:-1], N)
@@ -578,8 +701,11 @@ class LinearGeluFn(torch.autograd.Function):
             ),
             "sky_bias_gelu_bwd",
         )
-        dx = dpre.mm(weight).view(ctx.xshape)
-        dw = dpre.t().mm(x2)
+        dx = gemm2_dgrad(dpre, weight)
+        dx = dx.view(ctx.xshape) if dx is not None else dpre.mm(weight).view(ctx.xshape)
+        dw = gemm2_wgrad(dpre, x2)
+        if dw is None:
+            dw = dpre.t().mm(x2)
         return dx, dw, db
 
 
