@@ -510,8 +510,22 @@ def _g2_sites() -> set:
     return _G2_SITES
 
 
+_G2_SHAPES: set | None = None
+
+
+def _g2_shape_ok(M: int, N: int, K: int) -> bool:
+    """Optional per-shape allowlist: SKY_GEMM2_SHAPES="MxNxK,MxNxK"."""
+    global _G2_SHAPES
+    if _G2_SHAPES is None:
+        v = os.environ.get("SKY_GEMM2_SHAPES", "").strip()
+        _G2_SHAPES = ({tuple(int(d) for d in s.split("x")) for s in v.split(",") if s}
+                      if v else set())
+    return not _G2_SHAPES or (M, N, K) in _G2_SHAPES
+
+
 def _g2_fit(M: int, N: int, K: int) -> bool:
-    return M % 256 == 0 and N % 256 == 0 and K % 64 == 0
+    return (M % 256 == 0 and N % 256 == 0 and K % 64 == 0
+            and _g2_shape_ok(M, N, K))
 
 
 def _g2_gsu(M: int, N: int, K: int) -> int:

@@ -262,6 +262,10 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
     __builtin_amdgcn_s_setprio(0);
   };
 
+  // nounroll: a x2-unrolled body gets per-copy acc register assignments and
+  // the compiler bridges them with ~256 v_mov per iteration (measured in
+  // the asm); a single body keeps acc pinned.
+#pragma clang loop unroll(disable)
   for (int t = 0; t < NT; ++t) {
     char* abuf = lds + (t & 1) * 65536;
     char* bbuf = abuf + 32768;

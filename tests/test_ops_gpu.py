@@ -551,7 +551,7 @@ def test_gemm2_linear_fwd_bwd_matches_reference():
         torch.manual_seed(31)
         M, K, N = 4096, 1024, 3072
         x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda", requires_grad=True)
-        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda", requires_grad=True) * 0.03
+        w = (torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.03).requires_grad_()
         b = torch.randn(N, dtype=torch.bfloat16, device="cuda", requires_grad=True)
         y = LinearBiasFn.apply(x, w, b)
         dy = torch.randn_like(y) * 0.05
@@ -572,7 +572,7 @@ def test_gemm2_linear_fwd_bwd_matches_reference():
         # gelu-fused forward (bf16 path only exists through gemm2)
         M, K, N = 4096, 1024, 4096
         x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda", requires_grad=True)
-        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda", requires_grad=True) * 0.03
+        w = (torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.03).requires_grad_()
         b = torch.randn(N, dtype=torch.bfloat16, device="cuda", requires_grad=True)
         y = LinearGeluFn.apply(x, w, b)
         dy = torch.randn_like(y) * 0.05

@@ -18,6 +18,29 @@ import sys
 from collections import defaultdict
 
 
+def by_grid(db_path: str, flt: str = "") -> str:
+    """Per-(kernel, grid) means — disambiguates GEMM call sites that share
+    a kernel name but differ in problem shape (grid size)."""
+    db = sqlite3.connect(db_path)
+    cur = db.cursor()
+    tabs = [r[0] for r in cur.execute("select name from sqlite_master where type='table'")]
+    disp = next(t for t in tabs if t.startswith("rocpd_kernel_dispatch"))
+    sym = next(t for t in tabs if t.startswith("rocpd_info_kernel_symbol"))
+    names = {r[0]: r[1] for r in cur.execute(f"select id, display_name from {sym}")}
+    agg = defaultdict(lambda: [0.0, 0])
+    q = f"select kernel_id, grid_size_x, grid_size_y, grid_size_z, start, end from {disp}"
+    for kid, gx, gy, gz, s, e in cur.execute(q):
+        nm = names.get(kid, str(kid))
+        agg[(nm, gx, gy, gz)][0] += (e - s) / 1e6
+        agg[(nm, gx, gy, gz)][1] += 1
+    out = []
+    for (nm, gx, gy, gz), (ms, n) in sorted(agg.items(), key=lambda kv: -kv[1][0]):
+        if flt and flt.lower() not in nm.lower():
+            continue
+        out.append(f"{ms:9.2f}ms {n:6d}x {ms/n*1e3:8.2f}us  grid({gx},{gy},{gz})  {nm[:90]}")
+    return "\n".join(out[:60])
+
+
 def summarize(db_path: str, top: int = 40) -> str:
     db = sqlite3.connect(db_path)
     cur = db.cursor()
@@ -81,7 +104,9 @@ def diff(old_db: str, new_db: str, top: int = 40) -> str:
 
 
 if __name__ == "__main__":
-    if len(sys.argv) > 1 and sys.argv[1] == "--diff":
+    if len(sys.argv) > 1 and sys.argv[1] == "--by-grid":
+        print(by_grid(sys.argv[2], sys.argv[3] if len(sys.argv) > 3 else ""))
+    elif len(sys.argv) > 1 and sys.argv[1] == "--diff":
         print(diff(sys.argv[2], sys.argv[3],
                    int(sys.argv[4]) if len(sys.argv) > 4 else 40))
     else:
