@@ -142,6 +142,48 @@ class CommContext:
             works.append(_RetainingWork(dist.isend(buf, dst=dst), buf))
         return works
 
+    def irecv_tensors(self, src: int, key: str):
+        """Pre-post non-blocking receives for one microbatch hop so the
+        xGMI link fills while the current microbatch computes (SURVEY §2c
+        C4 overlap). Requires cached channel meta — returns None before the
+        first (handshaking) blocking recv on this channel."""
+        metas = self._meta_cache_recv.get((src, key))
+        if metas is None:
+            return None
+        bufs, works = [], []
+        for shape, dtype, req in metas:
+            b = torch.empty(shape, dtype=dtype, device=self.device)
+            works.append(dist.irecv(b, src=src))
+            bufs.append((b, req))
+        return bufs, works
+
+    def wait_irecv(self, posted) -> list[torch.Tensor]:
+        bufs, works = posted
+        for w in works:
+            w.wait()
+        out = []
+        for b, req in bufs:
+            if req:
+                b.requires_grad_(True)
+            out.append(b)
+        return out
+
+    def send_tensors_async(self, tensors: list, dst: int, key: str) -> list:
+        """Non-blocking sends when the channel meta is already cached (the
+        common steady state); falls back to the handshaking blocking send
+        on first use. Returns work handles to wait at phase end."""
+        metas = [
+            (tuple(t.shape), t.dtype, bool(t.requires_grad)) for t in tensors
+        ]
+        if self._meta_cache_send.get((dst, key)) != metas:
+            self.send_tensors(tensors, dst, key)
+            return []
+        works = []
+        for t in tensors:
+            buf = t.detach().contiguous()
+            works.append(_RetainingWork(dist.isend(buf, dst=dst), buf))
+        return works
+
     def recv_tensors_into(self, bufs: list, src: int):
         """Receive into preallocated (static) buffers — the graphed
         pipeline executor's transport (no allocation, no meta traffic)."""

@@ -28,6 +28,8 @@ the full-batch gradient (mean-reduction losses).
 
 from __future__ import annotations
 
+import os
+
 from dataclasses import dataclass, field
 
 import torch
@@ -204,6 +206,9 @@ class PipelineEngine:
 
         if schedule == "1f1b" and M > 1 and self.plan.num_stages > 1:
             total = self._run_1f1b(mb_inputs, mb_labels, M)
+        elif (schedule == "gpipe" and M > 1 and self.plan.num_stages > 1
+              and os.environ.get("SKY_NO_OVERLAP") != "1"):
+            total = self._run_gpipe_overlapped(mb_inputs, mb_labels, M)
         else:
             saved = []
             for m in range(M):
@@ -216,6 +221,68 @@ class PipelineEngine:
                     total += float(loss.detach())
             saved.clear()
         return self._broadcast_loss(total / M if self.is_last else None)
+
+    def _run_gpipe_overlapped(self, mb_inputs, mb_labels, M: int) -> float:
+        """GPipe with comm/compute overlap: the next
+        microbatch's activation/grad irecv is pre-posted while the current
+        one computes, and boundary sends are non-blocking (handles waited
+        at phase end). Falls back to blocking hops on the first iteration
+        of each channel (meta handshake)."""
+        send_works: list = []
+        saved = []
+        pending = None
+        for m in range(M):
+            if self.is_first:
+                args, held = _to_tuple(mb_inputs[m]), ()
+            else:
+                if pending is not None:
+                    args = tuple(self.comm.wait_irecv(pending))
+                else:
+                    args = tuple(self.comm.recv_tensors(self.prev_rank, "fwd"))
+                held = args
+                pending = (self.comm.irecv_tensors(self.prev_rank, "fwd")
+                           if m + 1 < M else None)
+            out = _to_tuple(self.stage(*args))
+            if not self.is_last:
+                send_works += self.comm.send_tensors_async(
+                    list(out), self.next_rank, "fwd")
+            saved.append((held, out))
+        for w in send_works:
+            w.wait()
+        send_works = []
+        total = 0.0
+        pending = None
+        for m in range(M):
+            held, out = saved[m]
+            if self.is_last:
+                logits = out[0] if len(out) == 1 else out
+                loss = self.loss_fn(logits, mb_labels[m].to(logits.device))
+                (loss / M).backward()
+                total += float(loss.detach())
+            else:
+                outs_req = [t for t in out if torch.is_tensor(t) and t.requires_grad]
+                if pending is not None:
+                    grads = self.comm.wait_irecv(pending)
+                else:
+                    grads = self.comm.recv_tensors(self.next_rank, "bwd")
+                assert len(grads) == len(outs_req), (
+                    f"stage {self.stage_idx}: got {len(grads)} grads for "
+                    f"{len(outs_req)} outputs")
+                pending = (self.comm.irecv_tensors(self.next_rank, "bwd")
+                           if m + 1 < M else None)
+                torch.autograd.backward(outs_req, grads)
+            if not self.is_first:
+                in_grads = [t.grad for t in held
+                            if torch.is_tensor(t) and t.requires_grad]
+                send_works += self.comm.send_tensors_async(
+                    in_grads, self.prev_rank, "bwd")
+                for t in held:
+                    if torch.is_tensor(t):
+                        t.grad = None
+            saved[m] = None
+        for w in send_works:
+            w.wait()
+        return total
 
     # ---------------- 1F1B ----------------
     # One-forward-one-backward: peak activation memory ~ num_stages

@@ -7,6 +7,7 @@ import subprocess
 import sys
 
 import numpy as np
+import pytest
 import torch
 
 from .helpers import run_multiprocess, tiny_bert_cfg
