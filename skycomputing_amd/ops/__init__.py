@@ -170,10 +170,10 @@ def linear_act(x, weight, bias, act: str = "gelu"):
         # v2 hand-GEMM path: one 256^2 8-phase kernel with the bias+gelu
         # epilogue fused (and the pre-activation stored for backward)
         if x.dtype == torch.bfloat16:
-            from .functions import _g2_fit, _g2_sites
+            from .functions import _g2_enabled, _g2_fit
 
             rows = x.numel() // x.shape[-1]
-            if ("fwd" in _g2_sites()
+            if (_g2_enabled("fwd", rows, weight.shape[0], weight.shape[1])
                     and _g2_fit(rows, weight.shape[0], weight.shape[1])):
                 return LinearGeluFn.apply(x, weight, bias)
         y = torch.nn.functional.linear(x, weight)
