@@ -133,6 +133,43 @@ __global__ __launch_bounds__(BLOCK) void bias_gelu_bwd_part_kernel(
   for (int j = 0; j < 8; ++j) scratch[(slab * cols8 + c8) * 8 + j] = s[j];
 }
 
+// Streaming fused dx + db partials (v2): unlike bias_gelu_bwd_part_kernel
+// (column-parallel, 2 KB-strided main traffic — measured SLOWER in-app,
+// r01), this keeps dx_vec's large sequential reads: each block owns a
+// (row-slab x 256-c8 column window) tile, reads/writes 4 KB strips per
+// row, and every THREAD keeps its fixed column-slice partial in registers
+// — db costs one fp32 slab write instead of a 32 MB re-read of dx.
+// Requires cols8 % 256 == 0 (cols % 2048 == 0; the FFN width).
+template <int DT, bool HAS_B>
+__global__ __launch_bounds__(256) void bias_gelu_bwd_cs_kernel(
+    const void* __restrict__ dy, const void* __restrict__ x,
+    const void* __restrict__ b, void* __restrict__ dx,
+    float* __restrict__ scratch, int64_t rows, int64_t cols8,
+    int64_t rows_per_slab) {
+  const int64_t nwin = cols8 >> 8;  // 256-c8 windows per row
+  const int64_t win = blockIdx.x % nwin;
+  const int64_t slab = blockIdx.x / nwin;
+  const int64_t c8 = win * 256 + threadIdx.x;
+  const int64_t r0 = slab * rows_per_slab;
+  const int64_t r1 = min(rows, r0 + rows_per_slab);
+  float bv[8];
+  if (HAS_B) Vec8<DT>::load(b, c8, bv);
+  float s[8] = {0.f};
+  for (int64_t r = r0; r < r1; ++r) {
+    float v[8], d[8];
+    Vec8<DT>::load(x, r * cols8 + c8, v);
+    Vec8<DT>::load(dy, r * cols8 + c8, d);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      d[j] *= gelu_grad_f(HAS_B ? v[j] + bv[j] : v[j]);
+      s[j] += d[j];
+    }
+    Vec8<DT>::store(dx, r * cols8 + c8, d);
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) scratch[(slab * cols8 + c8) * 8 + j] = s[j];
+}
+
 // db[c] = sum over rows of dx[r, c] — two-stage, no atomics, outputs need
 // no zero-init: stage 1 writes per-slab partials [nslabs][cols] fp32 to a
 // scratch buffer (thread owns 8 consecutive columns, 16 B loads); stage 2
@@ -265,6 +302,37 @@ SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
   // in-app (+4.9 us/layer vs the dx + two-stage-colsum chain at the bench
   // shape — profiles r01_bench160 diff); the proven chain is the default
   // and the fused kernel stays selectable for re-measurement.
+  // v2 streaming fused dx+db (see bias_gelu_bwd_cs_kernel); default where
+  // the shape allows, SKY_GELU_SPLIT_DB=1 restores the dx + colsum chain.
+  if (cols % 2048 == 0 && scratch != 0 && !getenv("SKY_GELU_SPLIT_DB")) {
+    const int64_t cols8 = cols / 8;
+    const int64_t nwin = cols8 >> 8;
+    int64_t nslabs = 2048 / nwin;
+    if (nslabs > CS_SLABS) nslabs = CS_SLABS;
+    if (nslabs > rows) nslabs = rows;
+    const int64_t slab = (rows + nslabs - 1) / nslabs;
+    dim3 grid((unsigned)(nwin * nslabs));
+    auto launch_cs = [&](auto kern) {
+      hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, (const void*)dy,
+                         (const void*)x, (const void*)b, (void*)dx,
+                         (float*)scratch, rows, cols8, slab);
+    };
+    if (dt == DT_F32)
+      hb ? launch_cs(bias_gelu_bwd_cs_kernel<DT_F32, true>)
+         : launch_cs(bias_gelu_bwd_cs_kernel<DT_F32, false>);
+    else
+      hb ? launch_cs(bias_gelu_bwd_cs_kernel<DT_BF16, true>)
+         : launch_cs(bias_gelu_bwd_cs_kernel<DT_BF16, false>);
+    dim3 g2((unsigned)((cols + 15) / 16));
+    if (dt == DT_BF16)
+      hipLaunchKernelGGL((colsum_final_kernel<256, DT_BF16>), g2, dim3(1024),
+                         0, s, (float*)scratch, (void*)db, cols, nslabs);
+    else
+      hipLaunchKernelGGL((colsum_final_kernel<256, DT_F32>), g2, dim3(1024),
+                         0, s, (float*)scratch, (void*)db, cols, nslabs);
+    LAUNCH_CHECK();
+    return 0;
+  }
   if (cols % 8 == 0 && scratch != 0 && getenv("SKY_GELU_FUSED_BWD")) {
     // fused one-pass dx + db partials (see bias_gelu_bwd_part_kernel).
     // This kernel moves 3 tensors (dy, x read; dx write), so unlike the

@@ -257,6 +257,112 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_vec_kernel(
   }
 }
 
+// Fused dx + dw/db column partials: the dx pass already holds dy and xhat
+// in registers for every element, and a wave's lane covers the SAME
+// column slices on every row it processes — so the dw/db partial sums
+// accumulate in registers for free and wb_part's 16-24 MB re-read of
+// dy/x disappears. Each WAVE is one scratch slab (slab = blockIdx*4+wid,
+// no LDS merge); the grid is capped so nslabs <= LN_SLABS.
+template <int DT, bool HAS_RES, bool DROP, int CH>
+__global__ __launch_bounds__(256) void ln_bwd_dx_cs_kernel(
+    const void* __restrict__ dy, const void* __restrict__ x,
+    const void* __restrict__ res, const void* __restrict__ w,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    void* __restrict__ dx, void* __restrict__ dres,
+    float* __restrict__ scratch, int64_t rows, int64_t cols, float keep,
+    uint64_t salt, const unsigned long long* __restrict__ state) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int64_t cols8 = cols / 8;
+  const uint64_t seed =
+      DROP ? salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull : 0;
+  const unsigned keep16 = keep_to_16(keep);
+  const float inv_keep = 1.f / keep;
+  float xh[CH][8], dxh[CH][8];
+  float sw[CH][8], sb[CH][8];
+#pragma unroll
+  for (int c = 0; c < CH; ++c)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { sw[c][j] = 0.f; sb[c][j] = 0.f; }
+  unsigned kbits[CH];
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wid; row < rows;
+       row += (int64_t)gridDim.x * 4) {
+    const int64_t base8 = row * cols8;
+    const float mu = mean[row], rs = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int c = 0; c < CH; ++c) {
+      const int64_t i8 = (int64_t)c * WAVE + lane;
+      if (i8 < cols8) {
+        float xv[8], dyv[8], wv[8];
+        Vec8<DT>::load(x, base8 + i8, xv);
+        if (DROP) {
+          const uint64_t e8 = (uint64_t)(base8 + i8);
+          const uint64_t z0 = rng_hash(seed, e8 * 2);
+          const uint64_t z1 = rng_hash(seed, e8 * 2 + 1);
+          kbits[c] = 0;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const uint64_t zz = j < 4 ? z0 : z1;
+            bool kb = (unsigned)((zz >> (16 * (j & 3))) & 0xFFFFu) < keep16;
+            kbits[c] |= (unsigned)kb << j;
+            xv[j] = kb ? xv[j] * inv_keep : 0.f;
+          }
+        }
+        if (HAS_RES) {
+          float r[8];
+          Vec8<DT>::load(res, base8 + i8, r);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) xv[j] += r[j];
+        }
+        Vec8<DT>::load(dy, base8 + i8, dyv);
+        Vec8<DT>::load(w, i8, wv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          xh[c][j] = (xv[j] - mu) * rs;
+          dxh[c][j] = dyv[j] * wv[j];
+          s1 += dxh[c][j];
+          s2 += dxh[c][j] * xh[c][j];
+          sw[c][j] += dyv[j] * xh[c][j];
+          sb[c][j] += dyv[j];
+        }
+      }
+    }
+    s1 = wave_sum(s1) / (float)cols;
+    s2 = wave_sum(s2) / (float)cols;
+#pragma unroll
+    for (int c = 0; c < CH; ++c) {
+      const int64_t i8 = (int64_t)c * WAVE + lane;
+      if (i8 < cols8) {
+        float o[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o[j] = rs * (dxh[c][j] - s1 - xh[c][j] * s2);
+        if (DROP && dres != nullptr) Vec8<DT>::store(dres, base8 + i8, o);
+        if (DROP) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            o[j] = ((kbits[c] >> j) & 1u) ? o[j] * inv_keep : 0.f;
+        }
+        Vec8<DT>::store(dx, base8 + i8, o);
+      }
+    }
+  }
+  // per-wave slab write (slabs that saw no rows still write their zeros)
+  float* base = scratch + ((int64_t)blockIdx.x * 4 + wid) * 2 * cols8 * 8;
+#pragma unroll
+  for (int c = 0; c < CH; ++c) {
+    const int64_t i8 = (int64_t)c * WAVE + lane;
+    if (i8 < cols8) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        base[i8 * 8 + j] = sw[c][j];
+        base[cols8 * 8 + i8 * 8 + j] = sb[c][j];
+      }
+    }
+  }
+}
+
 // column-parallel dw/db, two-stage, no atomics, outputs need no zero-init:
 // stage 1 writes per-slab partials to scratch [nslabs][2*cols] fp32
 // (dw partial at [y][c], db partial at [y][cols+c]); stage 2 reduces.
@@ -348,6 +454,38 @@ __global__ __launch_bounds__(1024) void ln_bwd_wb_final_kernel(
   }
 }
 
+// final reduction, CPB=4 columns per 1024-thread block (256 slab-groups):
+// 4x the block count and 4x the per-thread memory-level parallelism of the
+// 16-col variant (that one ran ~1 TB/s on an 8 MB scratch — latency-bound
+// at 64 blocks). Treats scratch as one [nslabs][2*cols] matrix; col >= cols
+// lands in db.
+template <int DTOUT>
+__global__ __launch_bounds__(1024) void ln_bwd_wb_final4_kernel(
+    const float* __restrict__ scratch, void* __restrict__ dw,
+    void* __restrict__ db, int64_t cols, int64_t nslabs) {
+  __shared__ float ldsv[256][5];
+  const int c = threadIdx.x & 3;
+  const int g = threadIdx.x >> 2;  // 0..255
+  const int64_t col = (int64_t)blockIdx.x * 4 + c;
+  const int64_t per = (nslabs + 255) / 256;
+  float s = 0.f;
+  if (col < 2 * cols) {
+    const int64_t y1 = min(nslabs, (int64_t)(g + 1) * per);
+    for (int64_t y = (int64_t)g * per; y < y1; ++y)
+      s += scratch[y * 2 * cols + col];
+  }
+  ldsv[g][c] = s;
+  __syncthreads();
+  for (int st = 128; st >= 1; st >>= 1) {
+    if (g < st) ldsv[g][c] += ldsv[g + st][c];
+    __syncthreads();
+  }
+  if (g == 0 && col < 2 * cols) {
+    if (col < cols) store_elem<DTOUT>(dw, col, ldsv[0][c]);
+    else store_elem<DTOUT>(db, col - cols, ldsv[0][c]);
+  }
+}
+
 template <int DT, bool HAS_RES, int BLOCK>
 __global__ __launch_bounds__(BLOCK) void ln_bwd_wb_kernel(
     const void* __restrict__ dy, const void* __restrict__ x,
@@ -412,6 +550,48 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
   bool has_res = res != 0;
   bool drop = keep < 1.f;
   if (drop && !ln_fast_ok(cols)) return (int)hipErrorInvalidValue;
+  // fused path: the dx kernel also produces the dw/db per-wave column
+  // partials (kills wb_part's re-read of dy/x); SKY_LN_SPLIT_WB=1 restores
+  // the separate 3-kernel chain for A/B.
+  const bool fused_wb =
+      ln_fast_ok(cols) && scratch != 0 && !getenv("SKY_LN_SPLIT_WB");
+  if (fused_wb) {
+    unsigned grid = (unsigned)((rows + 3) / 4);
+    if (grid > (unsigned)(LN_SLABS / 4)) grid = (unsigned)(LN_SLABS / 4);
+    const int ch = (int)((cols / 8 + WAVE - 1) / WAVE);
+#define LNCS(DT, HR, DR, CH)                                                   \
+  hipLaunchKernelGGL((ln_bwd_dx_cs_kernel<DT, HR, DR, CH>), dim3(grid),        \
+                     dim3(256), 0, s, (const void*)dy, (const void*)x,         \
+                     (const void*)res, (const void*)w, (const float*)mean,     \
+                     (const float*)rstd, (void*)dx, (void*)dres,               \
+                     (float*)scratch, rows, cols, keep, salt,                  \
+                     (const unsigned long long*)state)
+#define LNCS_CH(DT, HR, DR)                                                    \
+  do {                                                                         \
+    if (ch <= 2) LNCS(DT, HR, DR, 2);                                          \
+    else if (ch <= 4) LNCS(DT, HR, DR, 4);                                     \
+    else LNCS(DT, HR, DR, 8);                                                  \
+  } while (0)
+#define LNCS_D(DT, HR)                                                         \
+  do { if (drop) LNCS_CH(DT, HR, true); else LNCS_CH(DT, HR, false); } while (0)
+    if (dt == DT_F32) { if (has_res) LNCS_D(DT_F32, true); else LNCS_D(DT_F32, false); }
+    else              { if (has_res) LNCS_D(DT_BF16, true); else LNCS_D(DT_BF16, false); }
+#undef LNCS_D
+#undef LNCS_CH
+#undef LNCS
+    const int64_t nslabs = (int64_t)grid * 4;
+    dim3 g2((unsigned)((2 * cols + 3) / 4));
+    if (dt == DT_BF16)
+      hipLaunchKernelGGL((ln_bwd_wb_final4_kernel<DT_BF16>), g2, dim3(1024),
+                         0, s, (const float*)scratch, (void*)dw, (void*)db,
+                         cols, nslabs);
+    else
+      hipLaunchKernelGGL((ln_bwd_wb_final4_kernel<DT_F32>), g2, dim3(1024),
+                         0, s, (const float*)scratch, (void*)dw, (void*)db,
+                         cols, nslabs);
+    LAUNCH_CHECK();
+    return 0;
+  }
   if (ln_fast_ok(cols)) {
     unsigned grid = (unsigned)((rows + 3) / 4);
     if (grid > 8192u) grid = 8192u;
