@@ -307,7 +307,10 @@ SKY_EXPORT int sky_bias_gelu_bwd(uint64_t stream, uint64_t dy, uint64_t x,
   if (cols % 2048 == 0 && scratch != 0 && !getenv("SKY_GELU_SPLIT_DB")) {
     const int64_t cols8 = cols / 8;
     const int64_t nwin = cols8 >> 8;
-    int64_t nslabs = 2048 / nwin;
+    // ~1024 blocks feeds the streaming; fewer slabs shrink the final's
+    // scratch re-read (A/B'd: 1024 slabs put 16 MB on the final)
+    int64_t nslabs = 1024 / nwin;
+    if (const char* e = getenv("SKY_GELU_CS_SLABS")) nslabs = atoll(e);
     if (nslabs > CS_SLABS) nslabs = CS_SLABS;
     if (nslabs > rows) nslabs = rows;
     const int64_t slab = (rows + nslabs - 1) / nslabs;

@@ -348,18 +348,33 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_cs_kernel(
       }
     }
   }
-  // per-wave slab write (slabs that saw no rows still write their zeros)
-  float* base = scratch + ((int64_t)blockIdx.x * 4 + wid) * 2 * cols8 * 8;
+  // merge the 4 waves' partials through LDS (halves the slab count vs
+  // per-wave slabs -> half the scratch the final stage must re-read),
+  // then ONE slab write per block.
+  __shared__ float mrg[2][CH * 512];
+  for (int i = threadIdx.x; i < 2 * CH * 512; i += 256)
+    (&mrg[0][0])[i] = 0.f;
+  __syncthreads();
+  for (int k = 0; k < 4; ++k) {
+    if (wid == k) {
 #pragma unroll
-  for (int c = 0; c < CH; ++c) {
-    const int64_t i8 = (int64_t)c * WAVE + lane;
-    if (i8 < cols8) {
+      for (int c = 0; c < CH; ++c) {
+        const int64_t i8 = (int64_t)c * WAVE + lane;
+        if (i8 < cols8) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        base[i8 * 8 + j] = sw[c][j];
-        base[cols8 * 8 + i8 * 8 + j] = sb[c][j];
+          for (int j = 0; j < 8; ++j) {
+            mrg[0][i8 * 8 + j] += sw[c][j];
+            mrg[1][i8 * 8 + j] += sb[c][j];
+          }
+        }
       }
     }
+    __syncthreads();
+  }
+  float* base = scratch + (int64_t)blockIdx.x * 2 * cols8 * 8;
+  for (int i = threadIdx.x; i < cols8 * 8; i += 256) {
+    base[i] = mrg[0][i];
+    base[cols8 * 8 + i] = mrg[1][i];
   }
 }
 
@@ -556,8 +571,14 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
   const bool fused_wb =
       ln_fast_ok(cols) && scratch != 0 && !getenv("SKY_LN_SPLIT_WB");
   if (fused_wb) {
+    // one slab per BLOCK (LDS-merged): grid 512 keeps the dx streaming fed
+    // (256 starved it: 16.9 vs 9.9 us) while the final re-reads only
+    // 512 slabs (SKY_LN_CS_GRID overrides for sweeps)
+    unsigned gmax = 512;
+    if (const char* e = getenv("SKY_LN_CS_GRID")) gmax = (unsigned)atoi(e);
+    if (gmax > LN_SLABS) gmax = LN_SLABS;
     unsigned grid = (unsigned)((rows + 3) / 4);
-    if (grid > (unsigned)(LN_SLABS / 4)) grid = (unsigned)(LN_SLABS / 4);
+    if (grid > gmax) grid = gmax;
     const int ch = (int)((cols / 8 + WAVE - 1) / WAVE);
 #define LNCS(DT, HR, DR, CH)                                                   \
   hipLaunchKernelGGL((ln_bwd_dx_cs_kernel<DT, HR, DR, CH>), dim3(grid),        \
@@ -579,14 +600,14 @@ SKY_EXPORT int sky_layernorm_bwd(uint64_t stream, uint64_t dy, uint64_t x,
 #undef LNCS_D
 #undef LNCS_CH
 #undef LNCS
-    const int64_t nslabs = (int64_t)grid * 4;
-    dim3 g2((unsigned)((2 * cols + 3) / 4));
+    const int64_t nslabs = (int64_t)grid;  // one slab per block
+    dim3 g2((unsigned)((cols + 15) / 16));
     if (dt == DT_BF16)
-      hipLaunchKernelGGL((ln_bwd_wb_final4_kernel<DT_BF16>), g2, dim3(1024),
+      hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256, DT_BF16>), g2, dim3(1024),
                          0, s, (const float*)scratch, (void*)dw, (void*)db,
                          cols, nslabs);
     else
-      hipLaunchKernelGGL((ln_bwd_wb_final4_kernel<DT_F32>), g2, dim3(1024),
+      hipLaunchKernelGGL((ln_bwd_wb_final_kernel<256, DT_F32>), g2, dim3(1024),
                          0, s, (const float*)scratch, (void*)dw, (void*)db,
                          cols, nslabs);
     LAUNCH_CHECK();
