@@ -129,12 +129,31 @@ def attention(qkv, mask, dropout_p: float = 0.0, training: bool = False):
             out = FusedAttentionFn.apply(qkv, mask, scale, dropout_p, training)
             return out.reshape(B, S, h * d)
         if S <= 4096 and os.environ.get("SKY_FLASH_ATTN") == "1":
-            # flash-style forward (online softmax, no S x S tensor in
-            # forward). Currently a MEMORY option: the decomposed path
-            # below measures faster at these shapes (profiles/r01_notes.md),
-            # so it stays the default.
+            # our experimental flash-style forward (online softmax, no
+            # S x S tensor in forward; backward decomposes). Kept
+            # env-selectable; the sdpa path below beats it at every
+            # measured S (tools/flash_bench.py), so it is not the default.
             out = FlashAttentionFn.apply(qkv, mask, scale, dropout_p, training)
             return out.reshape(B, S, h * d)
+        # long-sequence default: torch-ROCm's flash scaled_dot_product
+        # _attention (fwd AND bwd tiled — no S x S tensor in either
+        # direction, unlike the decomposed path whose backward
+        # materializes scores). Same standing as hipBLASLt for plain
+        # GEMMs: library kernels where they win, hand kernels where ours
+        # do (S<=128 is the fused MFMA kernel above). Measured at S=512:
+        # sdpa ~26 us core vs 108 us decomposed / 176 us flash-v1.
+        if os.environ.get("SKY_NO_SDPA") != "1":
+            q = qkv[:, :, 0].permute(0, 2, 1, 3)
+            k = qkv[:, :, 1].permute(0, 2, 1, 3)
+            v = qkv[:, :, 2].permute(0, 2, 1, 3)
+            am = mask
+            if am is not None:
+                am = am.to(qkv.dtype)  # additive [B,1,1,S], broadcast
+            ctx = torch.nn.functional.scaled_dot_product_attention(
+                q, k, v, attn_mask=am,
+                dropout_p=dropout_p if training else 0.0,
+            )
+            return ctx.permute(0, 2, 1, 3).reshape(B, S, h * d)
     q = qkv[:, :, 0].permute(0, 2, 1, 3)
     k = qkv[:, :, 1].permute(0, 2, 1, 3)
     v = qkv[:, :, 2].permute(0, 2, 1, 3)

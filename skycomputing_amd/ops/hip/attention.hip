@@ -1018,7 +1018,21 @@ SKY_EXPORT int sky_attn_bwd(uint64_t stream, uint64_t qkv, uint64_t dout,
 // QK^T (two barriers per tile), per-wave P tile.
 // ============================================================================
 
-__global__ __launch_bounds__(256, 2) void attn_flash_fwd_kernel(
+// Flash forward v2: one workgroup covers QPW 128-query blocks of one
+// (batch, head) — K and V^T live in SEPARATE LDS regions staged ONCE per
+// kv tile and reused across the q blocks (the v1 kernel aliased V over K
+// and restaged both for every q block: S/128x K/V read amplification,
+// 176 vs 107 us at S=512 against the decomposed path). Consecutive
+// blockIdx values = same (b,h), so with the XCD-contiguous remap a head's
+// K/V streams from its XCD's L2 across its q-block workgroups.
+// LDS: K [128][64] @0 (16K), V^T [64][128] @16K, P per wave @32K+w*8K.
+#define F2_K_OFF 0
+#define F2_VT_OFF (16 * 1024)
+#define F2_P_OFF (32 * 1024)
+#define F2_LDS_BYTES (64 * 1024)
+
+template <int QPW>
+__global__ __launch_bounds__(256, 1) void attn_flash_fwd_kernel(
     const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ mask,
     ushort_t* __restrict__ out, float* __restrict__ m_io,
     float* __restrict__ l_io, int B, int S, int h, float scale, float keep,
@@ -1028,8 +1042,16 @@ __global__ __launch_bounds__(256, 2) void attn_flash_fwd_kernel(
   const int l = tid & 63;
   const int w = tid >> 6;
   const int nqb = (S + 127) / 128;
-  const int bh = blockIdx.x / nqb;
-  const int qb = blockIdx.x % nqb;
+  const int nchunk = (nqb + QPW - 1) / QPW;
+  int bid = blockIdx.x;
+  {  // XCD-contiguous remap (bijective): consecutive ids share an XCD
+    const int nwg = gridDim.x;
+    const int q = nwg / 8, rr = nwg % 8;
+    const int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int bh = bid / nchunk;
+  const int qc = bid % nchunk;
   const int b = bh / h;
   const int hh = bh % h;
   const int ts = 3 * h * ATT_D;
@@ -1043,41 +1065,42 @@ __global__ __launch_bounds__(256, 2) void attn_flash_fwd_kernel(
 
   const int lm = l & 15;
   const int lg = l >> 4;
-  const int q0 = qb * 128;          // this block's first query row
-  const int qt0 = 2 * w;            // wave's first q tile within the block
+  const int q0 = qc * QPW * 128;    // first query row of this chunk
+  const int qt0 = 2 * w;            // wave's first q tile within a block
 
-  // Q fragments for this wave's two q tiles (rows clamped)
-  bf16x8 aq_all[2][2];
+  // Q fragments for the wave's 2 tiles in each of the QPW blocks
+  bf16x8 aq_all[QPW][2][2];
 #pragma unroll
-  for (int qi = 0; qi < 2; ++qi) {
-    int qtok = q0 + (qt0 + qi) * 16 + lm;
-    if (qtok >= S) qtok = S - 1;
+  for (int blk = 0; blk < QPW; ++blk)
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks)
-      aq_all[qi][ks] = *(const bf16x8*)(qbase + (size_t)qtok * ts + ks * 32 + lg * 8);
-  }
-
-  // running state per (qi, r): max, sum; O accumulator
-  float mx[2][4], sm[2][4];
-  f32x4 oacc[2][4];
+    for (int qi = 0; qi < 2; ++qi) {
+      int qtok = q0 + blk * 128 + (qt0 + qi) * 16 + lm;
+      if (qtok >= S) qtok = S - 1;
 #pragma unroll
-  for (int qi = 0; qi < 2; ++qi)
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      mx[qi][r] = -3.0e38f;
-      sm[qi][r] = 0.f;
+      for (int ks = 0; ks < 2; ++ks)
+        aq_all[blk][qi][ks] =
+            *(const bf16x8*)(qbase + (size_t)qtok * ts + ks * 32 + lg * 8);
     }
+
+  float mx[QPW][2][4], sm[QPW][2][4];
+  f32x4 oacc[QPW][2][4];
 #pragma unroll
-  for (int qi = 0; qi < 2; ++qi)
+  for (int blk = 0; blk < QPW; ++blk)
 #pragma unroll
-    for (int dv = 0; dv < 4; ++dv) oacc[qi][dv] = (f32x4){0.f, 0.f, 0.f, 0.f};
+    for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        mx[blk][qi][r] = -3.0e38f;
+        sm[blk][qi][r] = 0.f;
+        oacc[blk][qi][r] = (f32x4){0.f, 0.f, 0.f, 0.f};
+      }
 
   const int nkv = (S + 127) / 128;
   for (int kv = 0; kv < nkv; ++kv) {
     const int k0 = kv * 128;
     const int ktiles = min(8, (S - k0 + 15) / 16);
+    __syncthreads();  // previous tile's K/Vt reads complete before overwrite
     // ---- stage K tile [128][64] via glds (source-side swizzle) ----
-    __syncthreads();  // previous tile's Vt reads complete before overwrite
     for (int u = tid; u < 128 * 8; u += 256) {
       const int tok = u >> 3;
       int gtok = k0 + tok;
@@ -1085,7 +1108,23 @@ __global__ __launch_bounds__(256, 2) void attn_flash_fwd_kernel(
       const int c16s = (u & 7) ^ (tok & 7);
       __builtin_amdgcn_global_load_lds(
           (att_gas)(kbase + (size_t)gtok * ts + c16s * 8),
-          (att_las)lds_at(lds, K_OFF + u * 16), 16, 0, 0);
+          (att_las)lds_at(lds, F2_K_OFF + u * 16), 16, 0, 0);
+    }
+    // ---- stage V^T [64 d][128 tok] (register transpose) ----
+    for (int u = tid; u < 128 * 8; u += 256) {
+      const int tok = u >> 3;
+      const int gtok = k0 + tok;
+      const int c16 = u & 7;
+      ushort8_t v;
+      if (gtok < S)
+        v = *(const ushort8_t*)(vbase + (size_t)gtok * ts + c16 * 8);
+      else
+        v = (ushort8_t)(ushort_t)0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int c = c16 * 8 + j;
+        *(ushort_t*)lds_at(lds, swz(F2_VT_OFF + c * 256 + tok * 2, c, 15)) = v[j];
+      }
     }
     // additive mask values for this tile's columns
     float mval[8];
@@ -1100,121 +1139,109 @@ __global__ __launch_bounds__(256, 2) void attn_flash_fwd_kernel(
     __syncthreads();
 
 #pragma clang loop unroll(disable)
-    for (int qi = 0; qi < 2; ++qi) {
-      const int qrow_base = q0 + (qt0 + qi) * 16;
-      if (qrow_base >= S) break;
-      f32x4 sacc[8];
-#pragma unroll
-      for (int kt = 0; kt < 8; ++kt) sacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kt = 0; kt < 8; ++kt) {
-        if (kt >= ktiles) continue;
-        const int ktok = kt * 16 + lm;
-#pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          bf16x8 bk = *(const bf16x8*)lds_at(
-              lds, swz(K_OFF + ktok * 128 + (ks * 32 + lg * 8) * 2, ktok, 7));
-          sacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aq_all[qi][ks], bk, sacc[kt], 0, 0, 0);
-        }
-      }
-      // online softmax update for this tile
-      float tmx[4];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) tmx[r] = -3.0e38f;
-#pragma unroll
-      for (int kt = 0; kt < 8; ++kt) {
-        if (kt >= ktiles) continue;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float zz = sacc[kt][r] * scale + mval[kt];
-          sacc[kt][r] = zz;
-          tmx[r] = fmaxf(tmx[r], zz);
-        }
-      }
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-#pragma unroll
-        for (int off = 8; off > 0; off >>= 1)
-          tmx[r] = fmaxf(tmx[r], __shfl_xor(tmx[r], off, 64));
-        const float mn = fmaxf(mx[qi][r], tmx[r]);
-        const float f = __expf(mx[qi][r] - mn);  // exp(-inf - mn) -> 0 ok
-        mx[qi][r] = mn;
-        sm[qi][r] *= f;
-        // rescale O rows r of this qi (row index = lg*4+r of each 16x16)
-#pragma unroll
-        for (int dv = 0; dv < 4; ++dv) oacc[qi][dv][r] *= f;
-      }
-      float tsum[4] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kt = 0; kt < 8; ++kt) {
-        if (kt >= ktiles) continue;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float p = __expf(sacc[kt][r] - mx[qi][r]);
-          sacc[kt][r] = p;
-          tsum[r] += p;
-        }
-      }
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-#pragma unroll
-        for (int off = 8; off > 0; off >>= 1) tsum[r] += __shfl_xor(tsum[r], off, 64);
-        sm[qi][r] += tsum[r];
-      }
-      // dropout + write UNNORMALIZED P tile to per-wave LDS (linear-index
-      // RNG so the generic dropout kernels can regenerate the mask)
-#pragma unroll
-      for (int kt = 0; kt < 8; ++kt) {
-        const int col = k0 + kt * 16 + lm;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = qrow_base + lg * 4 + r;
-          float p = (kt < ktiles) ? sacc[kt][r] : 0.f;
-          if (keep < 1.f && p != 0.f && row < S && col < S) {
-            const uint64_t idx = ((uint64_t)bh * S + row) * S + col;
-            p = rng_keep16(seed, idx, keep16) ? p * inv_keep : 0.f;
-          }
-          const int rl = qi * 16 + lg * 4 + r;
-          *(ushort_t*)lds_at(
-              lds, swz(P_OFF + w * 8192 + rl * 256 + (kt * 16 + lm) * 2, rl, 15)) =
-              f32_to_bf16((row < S && col < S) ? p : 0.f);
-        }
-      }
-    }
-
-    // ---- stage V^T over K's region, then accumulate PV ----
-    __syncthreads();  // K reads + P writes done
-    for (int u = tid; u < 128 * 8; u += 256) {
-      const int tok = u >> 3;
-      const int gtok = k0 + tok;
-      const int c16 = u & 7;
-      ushort8_t v;
-      if (gtok < S)
-        v = *(const ushort8_t*)(vbase + (size_t)gtok * ts + c16 * 8);
-      else
-        v = (ushort8_t)(ushort_t)0;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int c = c16 * 8 + j;
-        *(ushort_t*)lds_at(lds, swz(VT_OFF + c * 256 + tok * 2, c, 15)) = v[j];
-      }
-    }
-    __syncthreads();
+    for (int blk = 0; blk < QPW; ++blk) {
+      if (q0 + blk * 128 >= S) break;
 #pragma clang loop unroll(disable)
-    for (int qi = 0; qi < 2; ++qi) {
-      if (q0 + (qt0 + qi) * 16 >= S) break;
+      for (int qi = 0; qi < 2; ++qi) {
+        const int qrow_base = q0 + blk * 128 + (qt0 + qi) * 16;
+        if (qrow_base >= S) break;
+        f32x4 sacc[8];
 #pragma unroll
-      for (int ks = 0; ks < 4; ++ks) {
-        const int rl = qi * 16 + lm;
-        bf16x8 ap = *(const bf16x8*)lds_at(
-            lds, swz(P_OFF + w * 8192 + rl * 256 + (ks * 32 + lg * 8) * 2, rl, 15));
+        for (int kt = 0; kt < 8; ++kt) sacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int dvt = 0; dvt < 4; ++dvt) {
-          const int dv = dvt * 16 + lm;
-          bf16x8 bv = *(const bf16x8*)lds_at(
-              lds, swz(VT_OFF + dv * 256 + (ks * 32 + lg * 8) * 2, dv, 15));
-          oacc[qi][dvt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, oacc[qi][dvt], 0, 0, 0);
+        for (int kt = 0; kt < 8; ++kt) {
+          if (kt >= ktiles) continue;
+          const int ktok = kt * 16 + lm;
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks) {
+            bf16x8 bk = *(const bf16x8*)lds_at(
+                lds, swz(F2_K_OFF + ktok * 128 + (ks * 32 + lg * 8) * 2, ktok, 7));
+            sacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                aq_all[blk][qi][ks], bk, sacc[kt], 0, 0, 0);
+          }
+        }
+        // online softmax update for this tile
+        float tmx[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) tmx[r] = -3.0e38f;
+#pragma unroll
+        for (int kt = 0; kt < 8; ++kt) {
+          if (kt >= ktiles) continue;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float zz = sacc[kt][r] * scale + mval[kt];
+            sacc[kt][r] = zz;
+            tmx[r] = fmaxf(tmx[r], zz);
+          }
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+#pragma unroll
+          for (int off = 8; off > 0; off >>= 1)
+            tmx[r] = fmaxf(tmx[r], __shfl_xor(tmx[r], off, 64));
+          const float mn = fmaxf(mx[blk][qi][r], tmx[r]);
+          const float f = __expf(mx[blk][qi][r] - mn);
+          mx[blk][qi][r] = mn;
+          sm[blk][qi][r] *= f;
+#pragma unroll
+          for (int dv = 0; dv < 4; ++dv) oacc[blk][qi][dv][r] *= f;
+        }
+        float tsum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kt = 0; kt < 8; ++kt) {
+          if (kt >= ktiles) continue;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float p = __expf(sacc[kt][r] - mx[blk][qi][r]);
+            sacc[kt][r] = p;
+            tsum[r] += p;
+          }
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+#pragma unroll
+          for (int off = 8; off > 0; off >>= 1)
+            tsum[r] += __shfl_xor(tsum[r], off, 64);
+          sm[blk][qi][r] += tsum[r];
+        }
+        // dropout + UNNORMALIZED P tile to per-wave LDS (linear-index RNG)
+#pragma unroll
+        for (int kt = 0; kt < 8; ++kt) {
+          const int col = k0 + kt * 16 + lm;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int row = qrow_base + lg * 4 + r;
+            float p = (kt < ktiles) ? sacc[kt][r] : 0.f;
+            if (keep < 1.f && p != 0.f && row < S && col < S) {
+              const uint64_t idx = ((uint64_t)bh * S + row) * S + col;
+              p = rng_keep16(seed, idx, keep16) ? p * inv_keep : 0.f;
+            }
+            const int rl = qi * 16 + lg * 4 + r;
+            *(ushort_t*)lds_at(
+                lds,
+                swz(F2_P_OFF + w * 8192 + rl * 256 + (kt * 16 + lm) * 2, rl, 15)) =
+                f32_to_bf16((row < S && col < S) ? p : 0.f);
+          }
+        }
+        // ---- PV for this qi straight away (V^T resident, P per-wave) ----
+        // same-wave LDS RAW on the P tile: wait the ds_writes explicitly
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        {
+          const int rl = qi * 16 + lm;
+#pragma unroll
+          for (int ks = 0; ks < 4; ++ks) {
+            bf16x8 ap = *(const bf16x8*)lds_at(
+                lds, swz(F2_P_OFF + w * 8192 + rl * 256 + (ks * 32 + lg * 8) * 2,
+                         rl, 15));
+#pragma unroll
+            for (int dvt = 0; dvt < 4; ++dvt) {
+              const int dv = dvt * 16 + lm;
+              bf16x8 bv = *(const bf16x8*)lds_at(
+                  lds, swz(F2_VT_OFF + dv * 256 + (ks * 32 + lg * 8) * 2, dv, 15));
+              oacc[blk][qi][dvt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  ap, bv, oacc[blk][qi][dvt], 0, 0, 0);
+            }
+          }
         }
       }
     }
@@ -1222,23 +1249,26 @@ __global__ __launch_bounds__(256, 2) void attn_flash_fwd_kernel(
 
   // ---- epilogue: normalize by l, write out + stats ----
 #pragma unroll
-  for (int qi = 0; qi < 2; ++qi) {
-    const int qrow_base = q0 + (qt0 + qi) * 16;
-    if (qrow_base >= S) break;
+  for (int blk = 0; blk < QPW; ++blk) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int row = qrow_base + lg * 4 + r;
-      const float inv = 1.f / sm[qi][r];
-      if (lm == 0 && row < S) {
-        m_io[(size_t)bh * S + row] = mx[qi][r];
-        l_io[(size_t)bh * S + row] = sm[qi][r];
-      }
+    for (int qi = 0; qi < 2; ++qi) {
+      const int qrow_base = q0 + blk * 128 + (qt0 + qi) * 16;
+      if (qrow_base >= S) break;
 #pragma unroll
-      for (int dvt = 0; dvt < 4; ++dvt) {
-        const int dv = dvt * 16 + lm;
-        if (row < S)
-          out[(((size_t)b * S + row) * h + hh) * ATT_D + dv] =
-              f32_to_bf16(oacc[qi][dvt][r] * inv);
+      for (int r = 0; r < 4; ++r) {
+        const int row = qrow_base + lg * 4 + r;
+        const float inv = 1.f / sm[blk][qi][r];
+        if (lm == 0 && row < S) {
+          m_io[(size_t)bh * S + row] = mx[blk][qi][r];
+          l_io[(size_t)bh * S + row] = sm[blk][qi][r];
+        }
+#pragma unroll
+        for (int dvt = 0; dvt < 4; ++dvt) {
+          const int dv = dvt * 16 + lm;
+          if (row < S)
+            out[(((size_t)b * S + row) * h + hh) * ATT_D + dv] =
+                f32_to_bf16(oacc[blk][qi][dvt][r] * inv);
+        }
       }
     }
   }
@@ -1252,12 +1282,20 @@ SKY_EXPORT int sky_attn_flash_fwd(uint64_t stream, uint64_t qkv,
   if (d != ATT_D) return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
   const int nqb = (int)((S + 127) / 128);
-  dim3 grid((unsigned)(B * h * nqb));
-  hipLaunchKernelGGL(attn_flash_fwd_kernel, grid, dim3(256), ATT_LDS_BYTES, s,
-                     (const ushort_t*)qkv, (const ushort_t*)mask,
-                     (ushort_t*)out, (float*)m, (float*)lsum, (int)B, (int)S,
-                     (int)h, scale, keep, salt,
-                     (const unsigned long long*)state);
+  // 2 query blocks per workgroup where the grid stays >= 256 WGs
+  int qpw = (nqb >= 2 && B * h * ((nqb + 1) / 2) >= 512) ? 2 : 1;
+  const int nchunk = (nqb + qpw - 1) / qpw;
+  dim3 grid((unsigned)(B * h * nchunk));
+#define F2L(QPW)                                                               \
+  hipLaunchKernelGGL((attn_flash_fwd_kernel<QPW>), grid, dim3(256),            \
+                     F2_LDS_BYTES, s, (const ushort_t*)qkv,                    \
+                     (const ushort_t*)mask, (ushort_t*)out, (float*)m,         \
+                     (float*)lsum, (int)B, (int)S, (int)h, scale, keep, salt,  \
+                     (const unsigned long long*)state)
+  if (qpw == 2) F2L(2);
+  else F2L(1);
+#undef F2L
+
   LAUNCH_CHECK();
   return 0;
 }

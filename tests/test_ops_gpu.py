@@ -522,6 +522,37 @@ def test_flash_attention_long_seq(S):
     )
 
 
+@pytest.mark.parametrize("S", [512, 2048])
+def test_attention_sdpa_dispatch_long_seq(S):
+    """The DEFAULT long-S attention path (torch flash sdpa) fwd+bwd vs the
+    eager fp32 reference — the dispatch must have no S regression window
+    and no S x S tensor in either direction."""
+    torch.manual_seed(22)
+    from skycomputing_amd import ops
+
+    B, h, d = 2, 4, 64
+    qkv = torch.randn(B, S, 3, h, d, dtype=torch.bfloat16, device="cuda",
+                      requires_grad=True)
+    mask = torch.zeros(B, 1, 1, S, dtype=torch.bfloat16, device="cuda")
+    mask[:, :, :, S - S // 5:] = -10000.0
+    out = ops.attention(qkv, mask, 0.0, True)
+    assert out.shape == (B, S, h * d)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+
+    qf = qkv.detach().float().requires_grad_(True)
+    q = qf[:, :, 0].permute(0, 2, 1, 3)
+    k = qf[:, :, 1].permute(0, 2, 1, 3)
+    v = qf[:, :, 2].permute(0, 2, 1, 3)
+    ref = eager.attention_context(q, k, v, mask.float()).permute(0, 2, 1, 3)
+    ref = ref.reshape(B, S, h * d)
+    ref.backward(dout.float())
+    assert torch.allclose(out.float(), ref, atol=6e-2, rtol=6e-2), (
+        S, (out.float() - ref).abs().max())
+    assert torch.allclose(qkv.grad.float(), qf.grad, atol=8e-2, rtol=8e-2), (
+        S, (qkv.grad.float() - qf.grad).abs().max())
+
+
 def test_flash_attention_dropout_trains():
     torch.manual_seed(21)
     from skycomputing_amd.ops.functions import FlashAttentionFn
