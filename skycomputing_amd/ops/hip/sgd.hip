@@ -56,7 +56,7 @@ DEV void nt_store8<DT_F32>(void* p, int64_t i8, const float f[8]) {
   __builtin_nontemporal_store(b, (float4_t*)p + i8 * 2 + 1);
 }
 
-template <int DT, int BLOCK, bool MASTER, bool MOM>
+template <int DT, int BLOCK, bool MASTER, bool MOM, int UNROLL = 2>
 __global__ __launch_bounds__(BLOCK) void sgd_kernel(
     const SgdDesc* __restrict__ descs, float lr, float momentum, float wd) {
   const SgdDesc d = descs[blockIdx.x];
@@ -66,10 +66,10 @@ __global__ __launch_bounds__(BLOCK) void sgd_kernel(
   float* mbuf = (float*)d.mom;
   const int64_t n8 = d.n / 8;
   // vectorized body (slab bases are 16B-aligned; tail handled below);
-  // 2x unrolled for memory-level parallelism
-  for (int64_t i8 = threadIdx.x; i8 < n8; i8 += 2 * BLOCK) {
+  // UNROLLx for memory-level parallelism
+  for (int64_t i8 = threadIdx.x; i8 < n8; i8 += UNROLL * BLOCK) {
 #pragma unroll
-    for (int u = 0; u < 2; ++u) {
+    for (int u = 0; u < UNROLL; ++u) {
       const int64_t k8 = i8 + u * BLOCK;
       if (k8 >= n8) break;
       float gv[8], pv[8];
@@ -116,9 +116,23 @@ SKY_EXPORT int sky_sgd_step(uint64_t stream, uint64_t descs, int64_t n_descs,
   const bool has_mom = flags & 2;
   hipStream_t s = (hipStream_t)stream;
   dim3 grid((unsigned)n_descs);
+  const char* ue = getenv("SKY_SGD_UNROLL");
+  const int unroll = ue ? atoi(ue) : 2;
 #define SGD(DT, MA, MO)                                                      \
-  hipLaunchKernelGGL((sgd_kernel<DT, BLOCK, MA, MO>), grid, dim3(BLOCK), 0,  \
-                     s, (const SgdDesc*)descs, lr, momentum, wd)
+  do {                                                                       \
+    if (unroll >= 8)                                                         \
+      hipLaunchKernelGGL((sgd_kernel<DT, BLOCK, MA, MO, 8>), grid,           \
+                         dim3(BLOCK), 0, s, (const SgdDesc*)descs, lr,       \
+                         momentum, wd);                                      \
+    else if (unroll >= 4)                                                    \
+      hipLaunchKernelGGL((sgd_kernel<DT, BLOCK, MA, MO, 4>), grid,           \
+                         dim3(BLOCK), 0, s, (const SgdDesc*)descs, lr,       \
+                         momentum, wd);                                      \
+    else                                                                     \
+      hipLaunchKernelGGL((sgd_kernel<DT, BLOCK, MA, MO, 2>), grid,           \
+                         dim3(BLOCK), 0, s, (const SgdDesc*)descs, lr,       \
+                         momentum, wd);                                      \
+  } while (0)
   if (dt == DT_F32) {
     if (has_master) { if (has_mom) SGD(DT_F32, true, true); else SGD(DT_F32, true, false); }
     else            { if (has_mom) SGD(DT_F32, false, true); else SGD(DT_F32, false, false); }

@@ -9,12 +9,14 @@ steady state this is built once).
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 from . import hiplib
 from .hiplib import check
 
-_SLAB = 1 << 18  # elements per descriptor (smaller slabs -> more blocks, shorter serial chains)
+_SLAB = int(os.environ.get("SKY_SGD_SLAB", 1 << 20))  # elements per descriptor (1M-elem slabs measured fastest: 5.5 vs 4.7 TB/s at 256k, tools/sgd_bench.py)
 
 _PLAN_CACHE: dict[int, tuple] = {}
 
