@@ -706,7 +706,6 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1s_kernel(
   __syncthreads();
 
   // phase A: scores + dPd
-  ushort_t pd_keep[8][4];
   f32x4 sacc[8], dacc[8];
 #pragma unroll
   for (int kt = 0; kt < 8; ++kt) {
@@ -790,6 +789,7 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1s_kernel(
     for (int kt = 0; kt < 8; ++kt) {
       if (kt >= NT) continue;
       const int col = kt * 16 + lm;
+      ushort4_t ds4, pd4;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = qtok_base + lg * 4 + r;
@@ -806,29 +806,26 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1s_kernel(
         const int rl = lg * 4 + r;
         *(ushort_t*)lds_at(lds, swz(32768 + w * 4096 + rl * 256 + col * 2, rl, 15)) =
             f32_to_bf16(ds);
-        pd_keep[kt][r] = f32_to_bf16(pd);
+        ds4[r] = f32_to_bf16(ds);
+        pd4[r] = f32_to_bf16(pd);
+      }
+      if (col < S) {
+        const size_t o = ((size_t)bh * S + col) * S + qtok_base + lg * 4;
+        if (qtok_base + lg * 4 + 3 < S) {
+          *(ushort4_t*)(dsT + o) = ds4;
+          *(ushort4_t*)(pdT + o) = pd4;
+        } else {
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            if (qtok_base + lg * 4 + r < S) {
+              dsT[o + r] = ds4[r];
+              pdT[o + r] = pd4[r];
+            }
+        }
       }
     }
   }
-  __syncthreads();  // Kt + dS tiles complete (dQ MFMAs / pd writes next)
-
-  // pd tile into the DEAD K region (per wave [16 rows][128 cols] sw15) —
-  // the direct transposed global stores this replaces wrote 8-B pieces at
-  // 256-B stride (8-16x write amplification); the block-wide copy-out
-  // below writes full 16-B segments instead.
-  if (active) {
-#pragma unroll
-    for (int kt = 0; kt < 8; ++kt) {
-      if (kt >= NT) continue;
-      const int col = kt * 16 + lm;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int rl = lg * 4 + r;
-        *(ushort_t*)lds_at(lds, swz(16384 + w * 4096 + rl * 256 + col * 2, rl, 15)) =
-            pd_keep[kt][r];
-      }
-    }
-  }
+  __syncthreads();  // Kt complete before the dQ MFMAs read it
 
   if (active) {
     // dQ = dS K : A = this wave's dS rows (LDS), B = Kt
@@ -856,42 +853,6 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1s_kernel(
         const int row = qtok_base + lg * 4 + r;
         const int c = ct * 16 + lm;
         if (row < S) dq[(size_t)row * ts + c] = f32_to_bf16(qacc[ct][r]);
-      }
-    }
-  }
-  __syncthreads();  // pd tiles complete before the block-wide copy-out
-
-  // coalesced transposed copy-out: unit u = (col, row-octet); gathers 8
-  // row-values for one column from the four waves' LDS tiles and stores
-  // ONE contiguous 16-B segment of dsT/pdT (vs the old 8-B/256-B-stride
-  // scatter). The block covers 64 q rows, so segments tile exactly.
-  {
-    const int row0 = qb * 64;
-    for (int u = tid; u < 128 * 8; u += 256) {
-      const int col = u >> 3;
-      if (col >= S) break;
-      const int ro = u & 7;  // row octet
-      ushort8_t dsv, pdv;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int row = ro * 8 + j;           // block-local q row (0..63)
-        const int wv = row >> 4, rl = row & 15;
-        dsv[j] = *(const ushort_t*)lds_at(
-            lds, swz(32768 + wv * 4096 + rl * 256 + col * 2, rl, 15));
-        pdv[j] = *(const ushort_t*)lds_at(
-            lds, swz(16384 + wv * 4096 + rl * 256 + col * 2, rl, 15));
-      }
-      const size_t o = ((size_t)bh * S + col) * S + row0 + ro * 8;
-      if (row0 + ro * 8 + 8 <= S) {
-        *(ushort8_t*)(dsT + o) = dsv;
-        *(ushort8_t*)(pdT + o) = pdv;
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          if (row0 + ro * 8 + j < S) {
-            dsT[o + j] = dsv[j];
-            pdT[o + j] = pdv[j];
-          }
       }
     }
   }

@@ -16,7 +16,7 @@ import torch
 from . import hiplib
 from .hiplib import check
 
-_SLAB = int(os.environ.get("SKY_SGD_SLAB", 1 << 20))  # elements per descriptor (1M-elem slabs measured fastest: 5.5 vs 4.7 TB/s at 256k, tools/sgd_bench.py)
+_SLAB = int(os.environ.get("SKY_SGD_SLAB", 1 << 18))  # elements per descriptor (256k default: 1M-elem slabs win standalone (5.5 vs 4.7 TB/s) but LOSE in-app (+0.6 ms/step) - tools/sgd_bench.py + c7 profile)
 
 _PLAN_CACHE: dict[int, tuple] = {}
 
