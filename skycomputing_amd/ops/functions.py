@@ -308,12 +308,24 @@ class FusedAttentionFn(torch.autograd.Function):
         B, S, _, h, d = qkv.shape
         dev = qkv.device
         if os.environ.get("SKY_NO_FUSED_ATTN_BWD") != "1":
-            # fully-fused path: two MFMA kernels, no torch ops at all
             dout = dout.contiguous()
+            dqkv = torch.empty_like(qkv)
+            if os.environ.get("SKY_ATTN_SPLIT_BWD") != "1":
+                # single-kernel path: dQ/dK/dV in one launch, P and dS
+                # never leave LDS (no pdT/dsT scratch tensors at all)
+                check(
+                    lib.sky_attn_bwd_fused(
+                        _stream(), ptr(qkv), ptr(dout), ptr(mask), ptr(m),
+                        ptr(lsum), ptr(dqkv), B, S, h, d, ctx.scale,
+                        ctx.keep, ctx.salt, rng_state().data_ptr(),
+                    ),
+                    "sky_attn_bwd_fused",
+                )
+                return dqkv, None, None, None, None
+            # two-kernel path (bwd1s + bwd2 over transposed scratch)
             alloc = torch.empty if S == 128 else torch.zeros
             pdT = alloc((B, h, S, S), dtype=qkv.dtype, device=dev)
             dsT = alloc((B, h, S, S), dtype=qkv.dtype, device=dev)
-            dqkv = torch.empty_like(qkv)
             check(
                 lib.sky_attn_bwd(
                     _stream(), ptr(qkv), ptr(dout), ptr(mask), ptr(m),

@@ -947,6 +947,350 @@ __global__ __launch_bounds__(256) void attn_bwd2_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fully-fused attention backward (S <= 128): ONE kernel per (b,h) computes
+// dQ, dK, dV with P and dS living ONLY in LDS — the bwd1s/bwd2 pair's
+// transposed pdT/dsT scratch (2 x S x S bf16 per (b,h), written scattered
+// and re-read next kernel) never touches HBM, and Q/dO are staged once
+// instead of twice. Transposed operand fragments (Kt for dQ; pd^T/dS^T,
+// Q^T, dO^T for dK/dV) are read with ds_read_b64_tr_b16 (lane mapping
+// derived on HW, tools/tr16_probe.hip) straight from row-major images.
+//
+// LDS (80 KB -> 2 WGs/CU): K[128][64]sw7 @0, Q @16K, dO @32K (all glds-
+// staged), dS[64 q][128 k]sw15 @48K, pd @64K (per-pass tiles; q rows are
+// processed in two 64-row passes, dK/dV accumulate in registers across
+// passes; each wave owns one 16-row q tile per pass and k-tiles 2w,2w+1).
+#define ABF_K 0
+#define ABF_Q (16 * 1024)
+#define ABF_DO (32 * 1024)
+#define ABF_DS (48 * 1024)
+#define ABF_PD (64 * 1024)
+#define ABF_LDS (80 * 1024)
+
+typedef __attribute__((ext_vector_type(4))) short s16x4_a;
+typedef __attribute__((address_space(3))) s16x4_a* abf_las4;
+
+// tr16 read of an MFMA fragment from a row-major [row][64] sw7 image:
+// lane lm indexes the fragment's 16 output columns (cquad base cq0),
+// regs j = 4 consecutive image rows starting at r0 + (per-lane s>>2).
+DEV bf16x8 abf_tr_rows64(const char* lds_base, int off, int r0, int cq0,
+                         int s4) {
+  s16x4_a v[2];
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    const int row = r0 + half * 4 + (s4 >> 2);
+    const int byte = (off + row * 128 + (cq0 + (s4 & 3)) * 8) ^ ((row & 7) << 4);
+    v[half] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+        (abf_las4)(lds_base + byte));
+  }
+  return (bf16x8){v[0][0], v[0][1], v[0][2], v[0][3],
+                  v[1][0], v[1][1], v[1][2], v[1][3]};
+}
+
+// same for the [64 q][128 k] sw15 tiles (256-B rows)
+DEV bf16x8 abf_tr_rows128(const char* lds_base, int off, int r0, int cq0,
+                          int s4) {
+  s16x4_a v[2];
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    const int row = r0 + half * 4 + (s4 >> 2);
+    const int byte = (off + (row & 63) * 256 + (cq0 + (s4 & 3)) * 8) ^
+                     ((row & 15) << 4);
+    v[half] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+        (abf_las4)(lds_base + byte));
+  }
+  return (bf16x8){v[0][0], v[0][1], v[0][2], v[0][3],
+                  v[1][0], v[1][1], v[1][2], v[1][3]};
+}
+
+template <bool HAS_MASK>
+__global__ __launch_bounds__(256, 2) void attn_bwd_fused_kernel(
+    const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ dout,
+    const ushort_t* __restrict__ mask, const float* __restrict__ m_io,
+    const float* __restrict__ l_io, ushort_t* __restrict__ dqkv, int B,
+    int S, int h, float scale, float keep, uint64_t salt,
+    const unsigned long long* __restrict__ state) {
+  extern __shared__ __attribute__((aligned(16))) char lds[];
+  const int tid = threadIdx.x;
+  const int l = tid & 63;
+  const int w = tid >> 6;
+  const int lm = l & 15;
+  const int lg = l >> 4;
+  const int bh = blockIdx.x;
+  const int b = bh / h;
+  const int hh = bh % h;
+  const int ts = 3 * h * ATT_D;
+  const int dots = h * ATT_D;
+  const int NT = (S + 15) / 16;
+  const uint64_t seed = salt + (state ? *state : 0ull) * 0xD1B54A32D192ED03ull;
+  const float inv_keep = 1.f / keep;
+  const unsigned keep16 = keep_to_16(keep);
+
+  const ushort_t* qbase = qkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
+  const ushort_t* kbase = qbase + (size_t)h * ATT_D;
+  const ushort_t* vbase = qbase + (size_t)2 * h * ATT_D;
+  const ushort_t* dobase = dout + ((size_t)b * S * h + hh) * ATT_D;
+
+  // ---- stage K, Q, dO row images via glds (source-side swizzle) ----
+  for (int u = tid; u < 128 * 8; u += 256) {
+    const int tok = u >> 3;
+    const int gtok = tok < S ? tok : S - 1;
+    const int c16s = (u & 7) ^ (tok & 7);
+    __builtin_amdgcn_global_load_lds(
+        (att_gas)(kbase + (size_t)gtok * ts + c16s * 8),
+        (att_las)lds_at(lds, ABF_K + u * 16), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (att_gas)(qbase + (size_t)gtok * ts + c16s * 8),
+        (att_las)lds_at(lds, ABF_Q + u * 16), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (att_gas)(dobase + (size_t)gtok * dots + c16s * 8),
+        (att_las)lds_at(lds, ABF_DO + u * 16), 16, 0, 0);
+  }
+  // mask column values (pass-independent)
+  float mval[8];
+#pragma unroll
+  for (int kt = 0; kt < 8; ++kt) {
+    const int col = kt * 16 + lm;
+    float mv = 0.f;
+    if (HAS_MASK && col < S) mv = bf16_to_f32(mask[(size_t)b * S + col]);
+    mval[kt] = (col < S) ? mv : -3.0e38f;
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  // persistent dK/dV accumulators: wave w owns k-tiles 2w and 2w+1
+  f32x4 kacc[2][4], vacc[2][4];
+#pragma unroll
+  for (int ki = 0; ki < 2; ++ki)
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+      kacc[ki][ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
+      vacc[ki][ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
+    }
+
+  const int npass = (S + 63) / 64;
+  for (int pass = 0; pass < npass; ++pass) {
+    const int qtok_base = pass * 64 + w * 16;
+    const bool active = qtok_base < S;
+    // ---- phase A: scores (K from LDS) + dPd (V fragments from global);
+    // A fragments (this wave's q rows of Q / dO) read per-use from the
+    // staged LDS row images — no held registers ----
+    const int qrow = qtok_base + lm;
+    const int qrl = qrow < S ? qrow : S - 1;
+    f32x4 sacc[8], dacc[8];
+#pragma unroll
+    for (int kt = 0; kt < 8; ++kt) {
+      sacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
+      dacc[kt] = (f32x4){0.f, 0.f, 0.f, 0.f};
+    }
+    if (active) {
+      bf16x8 aq[2], ado[2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        aq[ks] = *(const bf16x8*)lds_at(
+            lds, swz(ABF_Q + qrl * 128 + (ks * 32 + lg * 8) * 2, qrl, 7));
+        ado[ks] = *(const bf16x8*)lds_at(
+            lds, swz(ABF_DO + qrl * 128 + (ks * 32 + lg * 8) * 2, qrl, 7));
+      }
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        if (kt >= NT) continue;
+        const int ktok = kt * 16 + lm;
+        const int vtok = ktok < S ? ktok : S - 1;
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          bf16x8 bk = *(const bf16x8*)lds_at(
+              lds, swz(ABF_K + ktok * 128 + (ks * 32 + lg * 8) * 2, ktok, 7));
+          sacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, sacc[kt], 0, 0, 0);
+          bf16x8 bv = *(const bf16x8*)(vbase + (size_t)vtok * ts + ks * 32 + lg * 8);
+          dacc[kt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[ks], bv, dacc[kt], 0, 0, 0);
+        }
+      }
+    }
+    // ---- P, dP, rowdot, dS/pd tiles (EVERY wave writes its 16 rows) ----
+    float mrow[4], lrow[4], dot[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qtok_base + lg * 4 + r;
+      const int rr = row < S ? row : S - 1;
+      mrow[r] = m_io[(size_t)bh * S + rr];
+      lrow[r] = 1.f / l_io[(size_t)bh * S + rr];
+      dot[r] = 0.f;
+    }
+    // dot pass: p and the dropout-masked dp are PARKED in the pd/dS tile
+    // slots (bf16) so sacc/dacc die here — the register peak would
+    // otherwise spill (sacc+dacc 64 + kacc/vacc 64 persistent).
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qtok_base + lg * 4 + r;
+      const int rl = lg * 4 + r;
+      uint64_t z0 = 0, z1 = 0;
+      if (keep < 1.f) {
+        const uint64_t zbase = ((uint64_t)bh * S + row) * 16 + lm;
+        z0 = rng_hash(seed, zbase * 2);
+        z1 = rng_hash(seed, zbase * 2 + 1);
+      }
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        const int col = kt * 16 + lm;
+        float p = (kt < NT)
+                      ? __expf(sacc[kt][r] * scale + mval[kt] - mrow[r]) * lrow[r]
+                      : 0.f;
+        if (!(row < S && col < S)) p = 0.f;
+        float dp = dacc[kt][r];
+        if (keep < 1.f) {
+          const uint64_t zz = kt < 4 ? z0 : z1;
+          bool kbit = (unsigned)((zz >> (16 * (kt & 3))) & 0xFFFFu) < keep16;
+          dp = kbit ? dp * inv_keep : 0.f;
+        }
+        dot[r] += dp * p;
+        *(ushort_t*)lds_at(lds, swz(ABF_PD + (w * 16 + rl) * 256 + col * 2, rl, 15)) =
+            f32_to_bf16(p);
+        *(ushort_t*)lds_at(lds, swz(ABF_DS + (w * 16 + rl) * 256 + col * 2, rl, 15)) =
+            f32_to_bf16(dp);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) dot[r] += __shfl_xor(dot[r], off, 64);
+    }
+    // finish pass: read the parked p/dp back, apply dropout to p -> pd,
+    // ds = scale * p * (dp - dot); overwrite the same slots in place.
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qtok_base + lg * 4 + r;
+      const int rl = lg * 4 + r;
+      uint64_t z0 = 0, z1 = 0;
+      if (keep < 1.f) {
+        const uint64_t zbase = ((uint64_t)bh * S + row) * 16 + lm;
+        z0 = rng_hash(seed, zbase * 2);
+        z1 = rng_hash(seed, zbase * 2 + 1);
+      }
+#pragma unroll
+      for (int kt = 0; kt < 8; ++kt) {
+        const int col = kt * 16 + lm;
+        const float p = bf16_to_f32(*(const ushort_t*)lds_at(
+            lds, swz(ABF_PD + (w * 16 + rl) * 256 + col * 2, rl, 15)));
+        const float dp = bf16_to_f32(*(const ushort_t*)lds_at(
+            lds, swz(ABF_DS + (w * 16 + rl) * 256 + col * 2, rl, 15)));
+        float pd = p;
+        if (keep < 1.f) {
+          const uint64_t zz = kt < 4 ? z0 : z1;
+          bool kbit = (unsigned)((zz >> (16 * (kt & 3))) & 0xFFFFu) < keep16;
+          pd = kbit ? p * inv_keep : 0.f;
+        }
+        const float ds = scale * p * (dp - dot[r]);
+        *(ushort_t*)lds_at(lds, swz(ABF_DS + (w * 16 + rl) * 256 + col * 2, rl, 15)) =
+            f32_to_bf16(ds);
+        *(ushort_t*)lds_at(lds, swz(ABF_PD + (w * 16 + rl) * 256 + col * 2, rl, 15)) =
+            f32_to_bf16(pd);
+      }
+    }
+    __syncthreads();  // tiles complete for cross-wave reads
+
+    // ---- dQ = dS K (A = own tile rows, direct b128; B = Kt via tr16) ----
+    if (active) {
+      f32x4 qacc[4];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) qacc[ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        if (ks * 32 >= S) break;
+        bf16x8 asr = *(const bf16x8*)lds_at(
+            lds, swz(ABF_DS + (w * 16 + lm) * 256 + (ks * 32 + lg * 8) * 2, lm, 15));
+#pragma unroll
+        for (int ct = 0; ct < 4; ++ct) {
+          bf16x8 bkt = abf_tr_rows64(lds, ABF_K, ks * 32 + lg * 8, ct * 4, lm);
+          qacc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(asr, bkt, qacc[ct], 0, 0, 0);
+        }
+      }
+      ushort_t* dq = dqkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = qtok_base + lg * 4 + r;
+          const int c = ct * 16 + lm;
+          if (row < S) dq[(size_t)row * ts + c] = f32_to_bf16(qacc[ct][r]);
+        }
+      }
+    }
+
+    // ---- dK += dS^T Q, dV += pd^T dO over this pass's 64 q rows ----
+    // wave's k-tiles: 2w, 2w+1; A frags via tr16 on the tiles, B frags
+    // via tr16 on the Q/dO row images (k-dim = global q row).
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      const int q0t = ks2 * 32 + lg * 8;             // tile-local q base
+      bf16x8 apd[2], ads[2];
+#pragma unroll
+      for (int ki = 0; ki < 2; ++ki) {
+        const int kt = 2 * w + ki;
+        apd[ki] = abf_tr_rows128(lds, ABF_PD, q0t, kt * 4, lm);
+        ads[ki] = abf_tr_rows128(lds, ABF_DS, q0t, kt * 4, lm);
+      }
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        bf16x8 bq = abf_tr_rows64(lds, ABF_Q, pass * 64 + q0t, ct * 4, lm);
+        bf16x8 bdo = abf_tr_rows64(lds, ABF_DO, pass * 64 + q0t, ct * 4, lm);
+#pragma unroll
+        for (int ki = 0; ki < 2; ++ki) {
+          vacc[ki][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(apd[ki], bdo, vacc[ki][ct], 0, 0, 0);
+          kacc[ki][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ads[ki], bq, kacc[ki][ct], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();  // tiles free for the next pass
+  }
+
+  // ---- write dK/dV (wave's 32 k rows) ----
+  ushort_t* dk = dqkv + (size_t)b * S * ts + (size_t)(h + hh) * ATT_D;
+  ushort_t* dv = dqkv + (size_t)b * S * ts + (size_t)(2 * h + hh) * ATT_D;
+#pragma unroll
+  for (int ki = 0; ki < 2; ++ki) {
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = (2 * w + ki) * 16 + lg * 4 + r;
+        const int c = ct * 16 + lm;
+        if (row < S) {
+          dk[(size_t)row * ts + c] = f32_to_bf16(kacc[ki][ct][r]);
+          dv[(size_t)row * ts + c] = f32_to_bf16(vacc[ki][ct][r]);
+        }
+      }
+    }
+  }
+}
+
+SKY_EXPORT int sky_attn_bwd_fused(uint64_t stream, uint64_t qkv,
+                                  uint64_t dout, uint64_t mask, uint64_t m,
+                                  uint64_t lsum, uint64_t dqkv, int64_t B,
+                                  int64_t S, int64_t h, int64_t d,
+                                  float scale, float keep, uint64_t salt,
+                                  uint64_t state) {
+  if (d != ATT_D || S > ATT_SMAX) return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid((unsigned)(B * h));
+  if (mask)
+    hipLaunchKernelGGL((attn_bwd_fused_kernel<true>), grid, dim3(256),
+                       ABF_LDS, s, (const ushort_t*)qkv,
+                       (const ushort_t*)dout, (const ushort_t*)mask,
+                       (const float*)m, (const float*)lsum, (ushort_t*)dqkv,
+                       (int)B, (int)S, (int)h, scale, keep, salt,
+                       (const unsigned long long*)state);
+  else
+    hipLaunchKernelGGL((attn_bwd_fused_kernel<false>), grid, dim3(256),
+                       ABF_LDS, s, (const ushort_t*)qkv,
+                       (const ushort_t*)dout, (const ushort_t*)mask,
+                       (const float*)m, (const float*)lsum, (ushort_t*)dqkv,
+                       (int)B, (int)S, (int)h, scale, keep, salt,
+                       (const unsigned long long*)state);
+  LAUNCH_CHECK();
+  return 0;
+}
+
 SKY_EXPORT int sky_attn_bwd(uint64_t stream, uint64_t qkv, uint64_t dout,
                             uint64_t mask, uint64_t m, uint64_t lsum,
                             uint64_t pdT, uint64_t dsT, uint64_t dqkv,

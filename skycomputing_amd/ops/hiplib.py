@@ -73,6 +73,9 @@ def _register_signatures(lib):
         #                      strm qkv mask out  m    l    B    S    h    d scale keep salt state
         "sky_attn_bwd": [u64] * 9 + [i64, i64, i64, i64, f32, f32, u64, u64],
         # strm qkv dout mask m l pdT dsT dqkv | B S h d scale keep salt state
+        "sky_attn_bwd_fused": [u64] * 7 + [i64, i64, i64, i64, f32, f32, u64, u64],
+        # strm qkv dout mask m l dqkv | B S h d scale keep salt state
+        # (ONE kernel: dQ/dK/dV, P+dS never leave LDS)
     }
     for name, argtypes in sigs.items():
         if hasattr(lib, name):
