@@ -1004,7 +1004,7 @@ DEV bf16x8 abf_tr_rows128(const char* lds_base, int off, int r0, int cq0,
 }
 
 template <bool HAS_MASK>
-__global__ __launch_bounds__(256, 2) void attn_bwd_fused_kernel(
+__global__ __launch_bounds__(256, 1) void attn_bwd_fused_kernel(
     const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ dout,
     const ushort_t* __restrict__ mask, const float* __restrict__ m_io,
     const float* __restrict__ l_io, ushort_t* __restrict__ dqkv, int B,
