@@ -49,3 +49,39 @@ for r in rows:
 print("\nkernel immediately before each copy (>=4KB):")
 for k, n in pred.most_common(12):
     print(f"{n:6d}  {k}")
+
+# --- copy/cast KERNELS (rocclr copyBuffer, at::copy casts): predecessors ---
+targets = ("copyBuffer", "bfloat16tofloat32", "bfloat16_copy", "float32tobfloat16")
+full = sorted(cur.execute(f"select start, kernel_id from {kd}"))
+names = [syms.get(k, "?") for _, k in full]
+pred2 = Counter()
+tot = Counter()
+for i, nm in enumerate(names):
+    for t in targets:
+        if t in nm:
+            tot[t] += 1
+            for j in range(i - 1, max(-1, i - 4), -1):
+                if all(x not in names[j] for x in targets):
+                    pred2[(t, names[j][:64])] += 1
+                    break
+            break
+print("\ncopy/cast kernel counts:", dict(tot))
+for (t, k), n in pred2.most_common(16):
+    print(f"{n:6d}  {t:22s} after {k}")
+
+# --- temporal: are copy/cast kernels setup-time or steady-state? ---
+all_start = full[0][0]
+all_end = max(s for s, _ in full)
+span = all_end - all_start
+buckets = Counter()
+for i, nm in enumerate(names):
+    for t in targets:
+        if t in nm:
+            frac = (full[i][0] - all_start) / span
+            buckets[(t, min(9, int(frac * 10)))] += 1
+            break
+print("\ntemporal deciles (0=start .. 9=end):")
+for t in targets:
+    row = [buckets.get((t, d), 0) for d in range(10)]
+    if sum(row):
+        print(f"{t:22s} {row}")
