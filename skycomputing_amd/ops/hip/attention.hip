@@ -48,7 +48,44 @@ DEV int swz(int byte_addr, int row, int rmask) {
   return byte_addr ^ ((row & rmask) << 4);
 }
 
-template <bool HAS_MASK, bool SAVE_ML, bool PROBS_MODE>
+typedef __attribute__((ext_vector_type(4))) short s16x4_a;
+typedef __attribute__((address_space(3))) s16x4_a* abf_las4;
+
+// tr16 read of an MFMA fragment from a row-major [row][64] sw7 image:
+// lane lm indexes the fragment's 16 output columns (cquad base cq0),
+// regs j = 4 consecutive image rows starting at r0 + (per-lane s>>2).
+DEV bf16x8 abf_tr_rows64(const char* lds_base, int off, int r0, int cq0,
+                         int s4) {
+  s16x4_a v[2];
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    const int row = r0 + half * 4 + (s4 >> 2);
+    const int byte = (off + row * 128 + (cq0 + (s4 & 3)) * 8) ^ ((row & 7) << 4);
+    v[half] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+        (abf_las4)(lds_base + byte));
+  }
+  return (bf16x8){v[0][0], v[0][1], v[0][2], v[0][3],
+                  v[1][0], v[1][1], v[1][2], v[1][3]};
+}
+
+// same for the [64 q][128 k] sw15 tiles (256-B rows)
+DEV bf16x8 abf_tr_rows128(const char* lds_base, int off, int r0, int cq0,
+                          int s4) {
+  s16x4_a v[2];
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    const int row = r0 + half * 4 + (s4 >> 2);
+    const int byte = (off + (row & 63) * 256 + (cq0 + (s4 & 3)) * 8) ^
+                     ((row & 15) << 4);
+    v[half] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+        (abf_las4)(lds_base + byte));
+  }
+  return (bf16x8){v[0][0], v[0][1], v[0][2], v[0][3],
+                  v[1][0], v[1][1], v[1][2], v[1][3]};
+}
+
+
+template <bool HAS_MASK, bool SAVE_ML, bool PROBS_MODE, bool TRV = false>
 __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const ushort_t* __restrict__ qkv, const ushort_t* __restrict__ mask,
     ushort_t* __restrict__ out, float* __restrict__ m_io,
@@ -250,22 +287,36 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 #pragma unroll
     for (int dv = 0; dv < 4; ++dv) oacc[qi][dv] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  // ---- stage V transposed over K's (now dead) region ----
+  // ---- stage V over K's (now dead) region ----
   __syncthreads();  // all waves done reading K and writing their P tiles
-  if (S < ATT_SMAX) {
-    for (int u = tid; u < (16 * 1024) / 16; u += 256)
-      *(ushort8_t*)lds_at(lds, VT_OFF + u * 16) = (ushort8_t)(ushort_t)0;
-    __syncthreads();  // zero before scattered staging writes land
-  }
-  for (int u = tid; u < S * 8; u += 256) {
-    const int tok = u >> 3;
-    const int c16 = u & 7;
-    ushort8_t v = *(const ushort8_t*)(vbase + (size_t)tok * ts + c16 * 8);
+  if (TRV) {
+    // DIRECT [tok][64] sw7 row image via glds; PV reads it TRANSPOSED
+    // with ds_read_b64_tr_b16 (garbage rows tok>=S multiply P zeros)
+    for (int u = tid; u < 128 * 8; u += 256) {
+      const int tok = u >> 3;
+      const int gtok = tok < S ? tok : S - 1;
+      const int c16s = (u & 7) ^ (tok & 7);
+      __builtin_amdgcn_global_load_lds(
+          (att_gas)(vbase + (size_t)gtok * ts + c16s * 8),
+          (att_las)lds_at(lds, VT_OFF + u * 16), 16, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  } else {
+    if (S < ATT_SMAX) {
+      for (int u = tid; u < (16 * 1024) / 16; u += 256)
+        *(ushort8_t*)lds_at(lds, VT_OFF + u * 16) = (ushort8_t)(ushort_t)0;
+      __syncthreads();  // zero before scattered staging writes land
+    }
+    for (int u = tid; u < S * 8; u += 256) {
+      const int tok = u >> 3;
+      const int c16 = u & 7;
+      ushort8_t v = *(const ushort8_t*)(vbase + (size_t)tok * ts + c16 * 8);
 #pragma unroll
-    for (int jj = 0; jj < 8; ++jj) {
-      const int j = (jj + tok) & 7;  // bank-spread write order (see bwd2)
-      const int c = c16 * 8 + j;
-      *(ushort_t*)lds_at(lds, swz(VT_OFF + c * 256 + tok * 2, c, 15)) = v[j];
+      for (int jj = 0; jj < 8; ++jj) {
+        const int j = (jj + tok) & 7;  // bank-spread write order (see bwd2)
+        const int c = c16 * 8 + j;
+        *(ushort_t*)lds_at(lds, swz(VT_OFF + c * 256 + tok * 2, c, 15)) = v[j];
+      }
     }
   }
   __syncthreads();
@@ -283,9 +334,14 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
           lds, swz(P_OFF + w * 8192 + rl * 256 + (ks * 32 + lg * 8) * 2, rl, 15));
 #pragma unroll
       for (int dvt = 0; dvt < 4; ++dvt) {
-        const int dv = dvt * 16 + lm;
-        bf16x8 bv = *(const bf16x8*)lds_at(
-            lds, swz(VT_OFF + dv * 256 + (ks * 32 + lg * 8) * 2, dv, 15));
+        bf16x8 bv;
+        if (TRV) {
+          bv = abf_tr_rows64(lds, VT_OFF, ks * 32 + lg * 8, dvt * 4, lm);
+        } else {
+          const int dv = dvt * 16 + lm;
+          bv = *(const bf16x8*)lds_at(
+              lds, swz(VT_OFF + dv * 256 + (ks * 32 + lg * 8) * 2, dv, 15));
+        }
         oacc[qi][dvt] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, oacc[qi][dvt], 0, 0, 0);
       }
@@ -369,13 +425,29 @@ static int attn_launch(uint64_t stream, uint64_t qkv, uint64_t mask,
   // SLOWER than this one-block kernel: 29.2 vs 27.2 us; the duplicated K
   // staging doesn't pay without an occupancy gain, and the fwd register
   // budget pins 2 waves/SIMD either way. See profiles/r01_notes.md.)
+  // Default: stage V DIRECT via glds and read PV fragments with
+  // ds_read_b64_tr_b16 — measured 27.0 vs 30.3 us standalone and
+  // -0.55 ms/step in-app over the register-transpose V^T staging.
+  // SKY_ATTN_TRV=0 restores the old staging for A/B.
+  const char* trve = getenv("SKY_ATTN_TRV");
+  const bool trv = !(trve && trve[0] == '0');
 #define ATT(HM, SML, PM)                                                      \
-  hipLaunchKernelGGL((attn_fwd_kernel<HM, SML, PM>), grid, dim3(256),         \
-                     lds_bytes, s, (const ushort_t*)qkv,                      \
-                     (const ushort_t*)mask, (ushort_t*)out, (float*)m,        \
-                     (float*)lsum, (ushort_t*)p, (ushort_t*)pd, (int)B,       \
-                     (int)S, (int)h, scale, keep, salt,                       \
-                     (const unsigned long long*)state)
+  do {                                                                        \
+    if (trv)                                                                  \
+      hipLaunchKernelGGL((attn_fwd_kernel<HM, SML, PM, true>), grid,          \
+                         dim3(256), lds_bytes, s, (const ushort_t*)qkv,       \
+                         (const ushort_t*)mask, (ushort_t*)out, (float*)m,    \
+                         (float*)lsum, (ushort_t*)p, (ushort_t*)pd, (int)B,   \
+                         (int)S, (int)h, scale, keep, salt,                   \
+                         (const unsigned long long*)state);                   \
+    else                                                                      \
+      hipLaunchKernelGGL((attn_fwd_kernel<HM, SML, PM, false>), grid,         \
+                         dim3(256), lds_bytes, s, (const ushort_t*)qkv,       \
+                         (const ushort_t*)mask, (ushort_t*)out, (float*)m,    \
+                         (float*)lsum, (ushort_t*)p, (ushort_t*)pd, (int)B,   \
+                         (int)S, (int)h, scale, keep, salt,                   \
+                         (const unsigned long long*)state);                   \
+  } while (0)
   if (probs_mode) { if (hm) ATT(true, false, true); else ATT(false, false, true); }
   else            { if (hm) ATT(true, true, false); else ATT(false, true, false); }
 #undef ATT
@@ -966,42 +1038,6 @@ __global__ __launch_bounds__(256) void attn_bwd2_kernel(
 #define ABF_DS (48 * 1024)
 #define ABF_PD (64 * 1024)
 #define ABF_LDS (80 * 1024)
-
-typedef __attribute__((ext_vector_type(4))) short s16x4_a;
-typedef __attribute__((address_space(3))) s16x4_a* abf_las4;
-
-// tr16 read of an MFMA fragment from a row-major [row][64] sw7 image:
-// lane lm indexes the fragment's 16 output columns (cquad base cq0),
-// regs j = 4 consecutive image rows starting at r0 + (per-lane s>>2).
-DEV bf16x8 abf_tr_rows64(const char* lds_base, int off, int r0, int cq0,
-                         int s4) {
-  s16x4_a v[2];
-#pragma unroll
-  for (int half = 0; half < 2; ++half) {
-    const int row = r0 + half * 4 + (s4 >> 2);
-    const int byte = (off + row * 128 + (cq0 + (s4 & 3)) * 8) ^ ((row & 7) << 4);
-    v[half] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-        (abf_las4)(lds_base + byte));
-  }
-  return (bf16x8){v[0][0], v[0][1], v[0][2], v[0][3],
-                  v[1][0], v[1][1], v[1][2], v[1][3]};
-}
-
-// same for the [64 q][128 k] sw15 tiles (256-B rows)
-DEV bf16x8 abf_tr_rows128(const char* lds_base, int off, int r0, int cq0,
-                          int s4) {
-  s16x4_a v[2];
-#pragma unroll
-  for (int half = 0; half < 2; ++half) {
-    const int row = r0 + half * 4 + (s4 >> 2);
-    const int byte = (off + (row & 63) * 256 + (cq0 + (s4 & 3)) * 8) ^
-                     ((row & 15) << 4);
-    v[half] = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-        (abf_las4)(lds_base + byte));
-  }
-  return (bf16x8){v[0][0], v[0][1], v[0][2], v[0][3],
-                  v[1][0], v[1][1], v[1][2], v[1][3]};
-}
 
 template <bool HAS_MASK>
 __global__ __launch_bounds__(256, 1) void attn_bwd_fused_kernel(
