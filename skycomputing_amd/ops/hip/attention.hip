@@ -800,20 +800,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1s_kernel(
       }
     }
   }
-  __syncthreads();  // V dead from here
-  // rebuild Kt [c][tok] sw15 over the V region (K beyond S is zeroed above)
-  for (int u = tid; u < 128 * 8; u += 256) {
-    const int tok = u >> 3;
-    const int c16 = u & 7;
-    ushort8_t kv = *(const ushort8_t*)lds_at(
-        lds, swz(16384 + tok * 128 + c16 * 16, tok, 7));
-#pragma unroll
-    for (int jj = 0; jj < 8; ++jj) {
-      const int j = (jj + tok) & 7;  // bank-spread write order (see bwd2)
-      const int c = c16 * 8 + j;
-      *(ushort_t*)lds_at(lds, swz(0 + c * 256 + tok * 2, c, 15)) = kv[j];
-    }
-  }
+  // (v2: no Kt rebuild — dQ reads the K image transposed via
+  // ds_read_b64_tr_b16, same recipe as the forward's V path)
 
   if (active) {
     // P, dP, rowdot, dS (p/pd recomputed in the store loop from live
@@ -897,10 +885,10 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1s_kernel(
       }
     }
   }
-  __syncthreads();  // Kt complete before the dQ MFMAs read it
+  __syncthreads();  // dS tiles complete before the dQ MFMAs read them
 
   if (active) {
-    // dQ = dS K : A = this wave's dS rows (LDS), B = Kt
+    // dQ = dS K : A = this wave's dS rows (LDS), B = K^T via tr16
     f32x4 qacc[4];
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct) qacc[ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
@@ -911,9 +899,7 @@ __global__ __launch_bounds__(256, 3) void attn_bwd1s_kernel(
           lds, swz(32768 + w * 4096 + lm * 256 + (ks * 32 + lg * 8) * 2, lm, 15));
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
-        const int c = ct * 16 + lm;
-        bf16x8 bkt = *(const bf16x8*)lds_at(
-            lds, swz(0 + c * 256 + (ks * 32 + lg * 8) * 2, c, 15));
+        bf16x8 bkt = abf_tr_rows64(lds, 16384, ks * 32 + lg * 8, ct * 4, lm);
         qacc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(asr, bkt, qacc[ct], 0, 0, 0);
       }
     }
@@ -950,26 +936,28 @@ __global__ __launch_bounds__(256) void attn_bwd2_kernel(
   const ushort_t* qbase = qkv + (size_t)b * S * ts + (size_t)hh * ATT_D;
   const ushort_t* dobase = dout + ((size_t)b * S * h + hh) * ATT_D;
 
-  if (S < ATT_SMAX) {
-    for (int u = tid; u < (32 * 1024) / 16; u += 256)
-      *(ushort8_t*)lds_at(lds, u * 16) = (ushort8_t)(ushort_t)0;
-    __syncthreads();
-  }
+  // Q and dO staged DIRECT [tok][64] sw7 via glds; the dV/dK B-fragments
+  // read them transposed with ds_read_b64_tr_b16 (v2 — replaces the
+  // register-scatter Qt/dOt build and its residual bank conflicts).
+  // A-side zeros (pdT/dsT rows beyond S) null any garbage tail rows.
   for (int u = tid; u < S * 8; u += 256) {
     const int tok = u >> 3;
-    const int c16 = u & 7;
-    ushort8_t qv = *(const ushort8_t*)(qbase + (size_t)tok * ts + c16 * 8);
-    ushort8_t dv = *(const ushort8_t*)(dobase + (size_t)tok * dots + c16 * 8);
-    // per-lane write order rotated by tok: simultaneous scatter stores
-    // otherwise share ~4 LDS banks across the wave (13% conflict in PMC)
-#pragma unroll
-    for (int jj = 0; jj < 8; ++jj) {
-      const int j = (jj + tok) & 7;
-      const int c = c16 * 8 + j;
-      *(ushort_t*)lds_at(lds, swz(0 + c * 256 + tok * 2, c, 15)) = qv[j];
-      *(ushort_t*)lds_at(lds, swz(16384 + c * 256 + tok * 2, c, 15)) = dv[j];
-    }
+    const int c16s = (u & 7) ^ (tok & 7);
+    __builtin_amdgcn_global_load_lds(
+        (att_gas)(qbase + (size_t)tok * ts + c16s * 8),
+        (att_las)lds_at(lds, 0 + u * 16), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (att_gas)(dobase + (size_t)tok * dots + c16s * 8),
+        (att_las)lds_at(lds, 16384 + u * 16), 16, 0, 0);
   }
+  // rows >= S must be ZERO: the pdT/dsT A-fragment reads of a partial
+  // 32-q block run past row S into the next k-row's data (row stride S),
+  // and only a zero B side nulls that contribution.
+  for (int u = S * 8 + tid; u < 128 * 8; u += 256) {
+    *(ushort8_t*)lds_at(lds, 0 + u * 16) = (ushort8_t)(ushort_t)0;
+    *(ushort8_t*)lds_at(lds, 16384 + u * 16) = (ushort8_t)(ushort_t)0;
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
   const int kt0 = 2 * w;  // this wave's two 16-row k tiles
@@ -993,12 +981,9 @@ __global__ __launch_bounds__(256) void attn_bwd2_kernel(
       bf16x8 ads = *(const bf16x8*)(dsT + abase);
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
-        const int c = ct * 16 + lm;
-        bf16x8 bdo = *(const bf16x8*)lds_at(
-            lds, swz(16384 + c * 256 + (ks * 32 + lg * 8) * 2, c, 15));
+        bf16x8 bdo = abf_tr_rows64(lds, 16384, ks * 32 + lg * 8, ct * 4, lm);
         vacc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(apd, bdo, vacc[ct], 0, 0, 0);
-        bf16x8 bq = *(const bf16x8*)lds_at(
-            lds, swz(0 + c * 256 + (ks * 32 + lg * 8) * 2, c, 15));
+        bf16x8 bq = abf_tr_rows64(lds, 0, ks * 32 + lg * 8, ct * 4, lm);
         kacc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ads, bq, kacc[ct], 0, 0, 0);
       }
     }
