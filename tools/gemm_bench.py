@@ -78,6 +78,7 @@ def bench_v2():
         ("ffnup 4096x4096", 4096, 4096, 1024, (1, 2)),
         ("ffndn 4096x1024", 4096, 1024, 4096, (1, 2, 4, 8)),
         ("sq 4096x4096x4096", 4096, 4096, 4096, (1,)),
+        ("sq 8192x8192x8192", 8192, 8192, 8192, (1,)),
     ]
     for name, M, N, K, gsus in shapes:
         x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
