@@ -188,14 +188,18 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  // prologue: virtual phases -6..-1 of the steady-state schedule
+  // prologue: both LDS buffers staged (tiles 0 and 1), in steady-state
+  // issue order; the vmcnt(8) retires tile 0, leaving tile 1's 8 ops in
+  // flight — the same depth the post-MFMA issue schedule maintains.
   issueA(0, 0);
   issueB(0, 1);
   issueA(0, 1);
   issueB(0, 0);
   issueA(1, 0);
   issueB(1, 1);
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  issueA(1, 1);
+  issueB(1, 0);
+  asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
   bf16x8 af[4][2], bfr[2][2];
@@ -266,38 +270,45 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
   // the compiler bridges them with ~256 v_mov per iteration (measured in
   // the asm); a single body keeps acc pinned.
 #pragma clang loop unroll(disable)
+  // Post-MFMA prefetch issue: each phase's glds targets the slot whose
+  // last ds_read RETIRED at this phase's own lgkmcnt(0) — so all four of
+  // tile t+2's halves issue during tile t (one phase after the slot
+  // frees), keeping ~2 full tiles (8-16 glds) in flight instead of 2
+  // halves. The boundary wait relaxes to vmcnt(8): tile t+1's issues may
+  // stay outstanding; everything older (tile t+2's data... consumed next)
+  // has retired.
   for (int t = 0; t < NT; ++t) {
     char* abuf = lds + (t & 1) * 65536;
     char* bbuf = abuf + 32768;
-    // phase 0: quadrant (0,0); issue A1[t+1]
+    // phase 0: quadrant (0,0)
     readA(abuf, 0);
     readB(bbuf, 0);
-    issueA(t + 1, 1);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     mma(0, 0);
     __builtin_amdgcn_s_barrier();
-    // phase 1: quadrant (0,1); af reused; issue B0[t+1]
+    // phase 1: quadrant (0,1); af reused; A0[t]'s slot freed at p0
     readB(bbuf, 1);
-    issueB(t + 1, 0);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     mma(0, 1);
-    __builtin_amdgcn_s_barrier();
-    // phase 2: quadrant (1,1); bfr reused; issue A0[t+2]
-    readA(abuf, 1);
     issueA(t + 2, 0);
+    __builtin_amdgcn_s_barrier();
+    // phase 2: quadrant (1,1); bfr reused; B1[t]'s slot freed at p1
+    readA(abuf, 1);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     mma(1, 1);
-    __builtin_amdgcn_s_barrier();
-    // phase 3: quadrant (1,0); issue B1[t+2]; tile-boundary vmcnt
-    readB(bbuf, 0);
     issueB(t + 2, 1);
+    __builtin_amdgcn_s_barrier();
+    // phase 3: quadrant (1,0); A1/B0[t]'s slots free at this phase's lgkm
+    readB(bbuf, 0);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     mma(1, 0);
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    issueA(t + 2, 1);
+    issueB(t + 2, 0);
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
     __builtin_amdgcn_s_barrier();
   }
 
