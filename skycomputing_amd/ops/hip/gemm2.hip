@@ -253,18 +253,21 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
       }
     }
   };
-  auto mma = [&](int mh, int nh) {
+  auto mma_span = [&](int mh, int nh, int i0, int i1) {
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
+    for (int i = 0; i < 4; ++i) {
+      if (i < i0 || i >= i1) continue;
 #pragma unroll
       for (int j = 0; j < 2; ++j)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
           acc[mh * 4 + i][nh * 2 + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i][ks], bfr[j][ks], acc[mh * 4 + i][nh * 2 + j], 0, 0, 0);
+    }
     __builtin_amdgcn_s_setprio(0);
   };
+  auto mma = [&](int mh, int nh) { mma_span(mh, nh, 0, 4); };
 
   // nounroll: a x2-unrolled body gets per-copy acc register assignments and
   // the compiler bridges them with ~256 v_mov per iteration (measured in
@@ -291,22 +294,25 @@ __global__ __launch_bounds__(G2_BLOCK, 1) void gemm2_kernel(
     readB(bbuf, 1);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    mma(0, 1);
-    issueA(t + 2, 0);
+    mma_span(0, 1, 0, 2);
+    issueA(t + 2, 0);   // DMA launches under the second MFMA half
+    mma_span(0, 1, 2, 4);
     __builtin_amdgcn_s_barrier();
     // phase 2: quadrant (1,1); bfr reused; B1[t]'s slot freed at p1
     readA(abuf, 1);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    mma(1, 1);
+    mma_span(1, 1, 0, 2);
     issueB(t + 2, 1);
+    mma_span(1, 1, 2, 4);
     __builtin_amdgcn_s_barrier();
     // phase 3: quadrant (1,0); A1/B0[t]'s slots free at this phase's lgkm
     readB(bbuf, 0);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    mma(1, 0);
+    mma_span(1, 0, 0, 2);
     issueA(t + 2, 1);
+    mma_span(1, 0, 2, 4);
     issueB(t + 2, 0);
     asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
     __builtin_amdgcn_s_barrier();
