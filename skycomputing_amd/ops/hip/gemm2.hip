@@ -414,6 +414,8 @@ SKY_EXPORT int sky_gemm2(uint64_t stream, uint64_t A, uint64_t B, uint64_t C,
                          int gsu) {
   if (M % G2_BM || N % G2_BN || K % G2_BK) return (int)hipErrorInvalidValue;
   if (gsu > 1 && (K % (gsu * G2_BK) || !Wk)) return (int)hipErrorInvalidValue;
+  // 16-B vector I/O: every row stride must keep 8-element alignment
+  if (lda % 8 || ldb % 8 || ldc % 8) return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
   const long ntiles = (M / G2_BM) * (N / G2_BN);
   dim3 grid((unsigned)(ntiles * (gsu > 1 ? gsu : 1)));
