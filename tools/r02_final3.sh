@@ -2,6 +2,8 @@
 set -x
 export TMPDIR=/tmp
 cd /root/repo
+rocm-smi --showclocks --showpower --showtemp > gpurun_out/f4_smi.log 2>&1 || true
+grep -iE "sclk|power|temp|junction" gpurun_out/f4_smi.log | head -6
 timeout 900 python -m pytest tests -m gpu -q > gpurun_out/f4_gputests.log 2>&1
 echo "gputests rc=$?"; tail -2 gpurun_out/f4_gputests.log
 for i in 1 2 3; do
