@@ -210,12 +210,15 @@ def test_mfma_layout():
 
 @pytest.mark.parametrize("S", [128, 100, 64, 48])
 @pytest.mark.parametrize("with_mask", [True, False])
-@pytest.mark.parametrize("bwd", ["split", "fused"])
+@pytest.mark.parametrize("bwd", ["split", "fused", "oldv"])
 def test_fused_attention_fwd_bwd(S, with_mask, bwd, monkeypatch):
     # bwd="fused" exercises the single-kernel dQ/dK/dV path
-    # (sky_attn_bwd_fused); "split" the default bwd1s+bwd2 pair.
+    # (sky_attn_bwd_fused); "split" the default bwd1s+bwd2 pair;
+    # "oldv" the register-transpose V-staging forward (SKY_ATTN_TRV=0).
     if bwd == "fused":
         monkeypatch.setenv("SKY_ATTN_FUSED_BWD", "1")
+    elif bwd == "oldv":
+        monkeypatch.setenv("SKY_ATTN_TRV", "0")
     torch.manual_seed(8)
     from skycomputing_amd.ops.functions import FusedAttentionFn
 
