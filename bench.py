@@ -68,7 +68,12 @@ def main():
     )
     torch.manual_seed(1234 + rank)
 
-    M = args.microbatches or (1 if world == 1 else 8)
+    # Default microbatch count: measured at N=1, total work W(M) grows
+    # steeply as microbatches shrink the GEMM/attention shapes (bsz 32:
+    # W(1,2,4,8) = 110/166/253/439 ms) — the bubble model
+    # step = (W(M)/N) * (M+N-1)/M then favors M ~= N/2 over the naive
+    # M=N (predicted N=8: 87 ms at M=4 vs 103 at M=8).
+    M = args.microbatches or (1 if world == 1 else max(1, world // 2))
     assert args.batch % M == 0
 
     bert_cfg = dict(
