@@ -9,7 +9,7 @@ Run directly (N=1) or under torch.distributed.run with --nproc-per-node N
 Extra knobs beyond the driver contract:
   --layers L            encoder layers (default 160)
   --allocate MODE       even | dynamic | optimal (default even)
-  --microbatches M      pipeline microbatches (default: 1 if N==1 else 8)
+  --microbatches M      pipeline microbatches (default: 1 if N==1 else N/2)
   --slowdowns CSV       per-rank injected compute slowdown factors
   --stimulate           seeded synthetic heterogeneity for the allocator
 """

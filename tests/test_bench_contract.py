@@ -74,7 +74,7 @@ def test_bench_8rank_driver_invocation(tmp_path):
     d = json.loads(json_lines[0])
     assert d["n_gpus"] == 8
     assert d["config"]["parallelism"] == "pp8"
-    assert d["config"]["microbatches"] == 8
+    assert d["config"]["microbatches"] == 4  # default M = N/2 (measured W(M) curve)
     assert d["value"] > 0
 
 
