@@ -78,6 +78,19 @@ def test_bench_8rank_driver_invocation(tmp_path):
     assert d["value"] > 0
 
 
+def test_bench_2rank_default_microbatches(tmp_path):
+    """N=2 with driver-default flags: default M = N/2 = 1 (sequential
+    relay — the measured W(M) optimum at this batch)."""
+    out = tmp_path / "b2.json"
+    r = _torchrun_bench(2, ["--gpus", "2", "--layers", "4", "--batch", "8",
+                            "--seq", "16", "--steps", "2", "--warmup", "1"], out)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][0])
+    assert d["config"]["parallelism"] == "pp2"
+    assert d["config"]["microbatches"] == 1
+    assert d["value"] > 0
+
+
 def test_bench_virtual_stages_cpu(tmp_path):
     """Opt-in interleaved virtual stages through the bench entrypoint
     (4 ranks x 2 chunks on gloo/CPU)."""
