@@ -2,7 +2,7 @@ from setuptools import find_packages, setup
 
 setup(
     name="skycomputing_amd",
-    version="0.1.0",
+    version="0.2.0",
     description=(
         "MI355X-native load-balanced pipeline-parallel training framework "
         "(capabilities of hpcaitech/SkyComputing, rebuilt for gfx950/CDNA4)"

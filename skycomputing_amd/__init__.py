@@ -16,7 +16,7 @@ A from-scratch rebuild of the capabilities of hpcaitech/SkyComputing
 See SURVEY.md for the reference analysis this build follows.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from . import builder, config, dataset, dynamics, models, ops, parallel, registry, runner
 from .logger import Logger
