@@ -19,10 +19,11 @@ struct SgdDesc {
   int64_t pad;
 };
 
-// Every byte in the SGD step is touch-once (grads, masters, params are not
-// re-read until the next iteration), so all vector I/O is NON-TEMPORAL:
-// streaming loads/stores bypass L2 retention and avoid write-allocate
-// pollution, which is worth real bandwidth on an 8 TB/s HBM3E part.
+// NOTE on non-temporal I/O: every byte in the SGD step is touch-once, so
+// r01 used NT streaming loads/stores throughout. Measured in-app in r02,
+// PLAIN write-back I/O is 1.1 ms/step faster on the full bench — the NT
+// hints cost more at the memory controller than the avoided L2
+// write-allocate saves. NT kept selectable (SKY_SGD_NONT=0 -> NT).
 template <int DT>
 DEV void nt_load8(const void* p, int64_t i8, float f[8]);
 template <>
@@ -56,7 +57,8 @@ DEV void nt_store8<DT_F32>(void* p, int64_t i8, const float f[8]) {
   __builtin_nontemporal_store(b, (float4_t*)p + i8 * 2 + 1);
 }
 
-template <int DT, int BLOCK, bool MASTER, bool MOM, int UNROLL = 2>
+template <int DT, int BLOCK, bool MASTER, bool MOM, int UNROLL = 2,
+          bool NT = true>
 __global__ __launch_bounds__(BLOCK) void sgd_kernel(
     const SgdDesc* __restrict__ descs, float lr, float momentum, float wd) {
   const SgdDesc d = descs[blockIdx.x];
@@ -73,11 +75,15 @@ __global__ __launch_bounds__(BLOCK) void sgd_kernel(
       const int64_t k8 = i8 + u * BLOCK;
       if (k8 >= n8) break;
       float gv[8], pv[8];
-      nt_load8<DT>(g, k8, gv);
-      if (MASTER)
-        nt_load8<DT_F32>(master, k8, pv);
-      else
-        nt_load8<DT>(p, k8, pv);
+      if (NT) {
+        nt_load8<DT>(g, k8, gv);
+        if (MASTER) nt_load8<DT_F32>(master, k8, pv);
+        else nt_load8<DT>(p, k8, pv);
+      } else {
+        Vec8<DT>::load(g, k8, gv);
+        if (MASTER) Vec8<DT_F32>::load(master, k8, pv);
+        else Vec8<DT>::load(p, k8, pv);
+      }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float gg = gv[j];
@@ -89,8 +95,13 @@ __global__ __launch_bounds__(BLOCK) void sgd_kernel(
         }
         pv[j] -= lr * gg;
       }
-      if (MASTER) nt_store8<DT_F32>(master, k8, pv);
-      nt_store8<DT>(p, k8, pv);
+      if (NT) {
+        if (MASTER) nt_store8<DT_F32>(master, k8, pv);
+        nt_store8<DT>(p, k8, pv);
+      } else {
+        if (MASTER) Vec8<DT_F32>::store(master, k8, pv);
+        Vec8<DT>::store(p, k8, pv);
+      }
     }
   }
   for (int64_t i = n8 * 8 + threadIdx.x; i < d.n; i += BLOCK) {
@@ -118,9 +129,19 @@ SKY_EXPORT int sky_sgd_step(uint64_t stream, uint64_t descs, int64_t n_descs,
   dim3 grid((unsigned)n_descs);
   const char* ue = getenv("SKY_SGD_UNROLL");
   const int unroll = ue ? atoi(ue) : 2;
+  // plain (write-back) I/O is the measured in-app default: 108.95 vs
+  // 110.07 ms/step against the non-temporal variant (same box) — the r01
+  // touch-once/NT reasoning loses to L2 write buffering in practice.
+  // SKY_SGD_NONT=0 selects the NT variant for re-measurement.
+  const char* nte = getenv("SKY_SGD_NONT");
+  const bool plain = !(nte && nte[0] == '0');
 #define SGD(DT, MA, MO)                                                      \
   do {                                                                       \
-    if (unroll >= 8)                                                         \
+    if (plain)                                                               \
+      hipLaunchKernelGGL((sgd_kernel<DT, BLOCK, MA, MO, 2, false>), grid,    \
+                         dim3(BLOCK), 0, s, (const SgdDesc*)descs, lr,       \
+                         momentum, wd);                                      \
+    else if (unroll >= 8)                                                    \
       hipLaunchKernelGGL((sgd_kernel<DT, BLOCK, MA, MO, 8>), grid,           \
                          dim3(BLOCK), 0, s, (const SgdDesc*)descs, lr,       \
                          momentum, wd);                                      \
